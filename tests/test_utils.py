@@ -1,0 +1,101 @@
+import numpy as np
+import pytest
+import torch
+
+from ding.utils import (
+    EasyDict, SumSegmentTree, MinSegmentTree, deep_merge_dicts, lists_to_dicts, dicts_to_lists,
+    squeeze, set_pkg_seed, split_data_generator, Registry, RunningMeanStd, EasyTimer,
+    get_data_compressor, get_data_decompressor,
+)
+
+
+def test_easydict():
+    d = EasyDict({"a": 1, "b": {"c": [2, {"d": 3}]}})
+    assert d.a == 1 and d.b.c[1].d == 3
+    d.x = {"y": 5}
+    assert d.x.y == 5
+    import copy
+    d2 = copy.deepcopy(d)
+    d2.b.c[1].d = 9
+    assert d.b.c[1].d == 3
+
+
+def test_registry():
+    R = Registry("test")
+
+    @R.register("foo")
+    class Foo:
+        pass
+
+    assert R.get("foo") is Foo
+    assert "foo" in R
+    with pytest.raises(KeyError):
+        @R.register("foo")
+        class Bar:
+            pass
+
+
+def test_segment_tree_vs_numpy():
+    rng = np.random.RandomState(0)
+    cap = 128
+    st = SumSegmentTree(cap)
+    vals = rng.rand(cap)
+    st[np.arange(cap)] = vals
+    assert abs(st.reduce() - vals.sum()) < 1e-9
+    assert abs(st.reduce(10, 50) - vals[10:50].sum()) < 1e-9
+    # prefix-sum sampling correctness
+    for p in [0.0, vals.sum() * 0.3, vals.sum() * 0.999]:
+        i = st.find_prefixsum_idx(p)
+        cs = np.cumsum(vals)
+        expect = int(np.searchsorted(cs, p, side="right"))
+        assert i == expect
+    mt = MinSegmentTree(cap)
+    mt[np.arange(cap)] = vals
+    assert abs(mt.reduce(5, 77) - vals[5:77].min()) < 1e-12
+
+
+def test_dict_list_helpers():
+    data = [{"a": 1, "b": 2}, {"a": 3, "b": 4}]
+    d = lists_to_dicts(data)
+    assert d == {"a": [1, 3], "b": [2, 4]}
+    assert dicts_to_lists(d) == data
+    assert squeeze((4, )) == 4
+    merged = deep_merge_dicts({"a": {"b": 1, "c": 2}}, {"a": {"b": 7}})
+    assert merged == {"a": {"b": 7, "c": 2}}
+
+
+def test_split_data_generator():
+    data = {"obs": torch.arange(10).float(), "scalar": 3}
+    batches = list(split_data_generator(data, 5, shuffle=False))
+    assert len(batches) == 2 and batches[0]["obs"].shape[0] == 5
+    assert batches[0]["scalar"] == 3
+
+
+def test_running_mean_std():
+    rms = RunningMeanStd(shape=(3, ))
+    data = np.random.RandomState(1).randn(1000, 3) * 2 + 5
+    for i in range(0, 1000, 100):
+        rms.update(data[i:i + 100])
+    assert np.allclose(rms.mean, data.mean(0), atol=1e-2)
+    assert np.allclose(rms.std, data.std(0), atol=1e-2)
+
+
+def test_compression_roundtrip():
+    payload = {"x": np.arange(100), "y": [1, "two"]}
+    for name in ("none", "zlib", "lz4"):
+        c = get_data_compressor(name)(payload)
+        out = get_data_decompressor(name)(c)
+        assert out["y"] == payload["y"] and (out["x"] == payload["x"]).all()
+
+
+def test_timer():
+    with EasyTimer(cuda=False) as t:
+        sum(range(1000))
+    assert t.value >= 0
+
+
+def test_seed():
+    set_pkg_seed(42, use_cuda=False)
+    a = torch.rand(3)
+    set_pkg_seed(42, use_cuda=False)
+    assert torch.equal(a, torch.rand(3))
