@@ -85,6 +85,14 @@ def multistep_forward_view(bootstrap_values, rewards, gammas, lambda_, done):
     ).to(rewards.dtype)
 
 
+def scatter_connection(x, index, H: int, W: int, scatter_type: str):
+    """x [B,M,N], index [B,M] flat spatial positions -> [B,N,H,W]."""
+    ext = _load()
+    return ext.scatter_connection(
+        x.contiguous().float(), index.contiguous().long(), int(H), int(W), 1 if scatter_type == 'add' else 0
+    ).to(x.dtype)
+
+
 def c51_project(next_n_dist, next_n_act, reward_n, done, v_min, v_max, gamma_n):
     """Categorical projection of r + gamma_n * z onto the fixed support.
 
