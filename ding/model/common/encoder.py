@@ -1,0 +1,203 @@
+"""Observation encoders.
+
+Parity: reference ding/model/common/encoder.py (ConvEncoder:24, FCEncoder:158,
+IMPALAConvEncoder:390, GaussianFourierProjectionTimeEncoder:476).
+"""
+import math
+from typing import Dict, List, Optional, Union
+
+import torch
+import torch.nn as nn
+
+from ding.torch_utils import ResBlock, ResFCBlock, conv2d_block, fc_block, build_activation, normed_linear, normed_conv2d
+
+
+def prod(iterable):
+    out = 1
+    for x in iterable:
+        out *= x
+    return out
+
+
+class ConvEncoder(nn.Module):
+    """Nature-DQN style conv stack + flatten + fc to hidden_size_list[-1]."""
+
+    def __init__(
+        self,
+        obs_shape: tuple,
+        hidden_size_list: List[int] = [32, 64, 64, 128],
+        activation: str = 'relu',
+        kernel_size: List[int] = [8, 4, 3],
+        stride: List[int] = [4, 2, 1],
+        padding: Optional[List[int]] = None,
+        layer_norm: bool = False,
+        norm_type: Optional[str] = None,
+    ):
+        super().__init__()
+        self.obs_shape = obs_shape
+        if padding is None:
+            padding = [0] * len(kernel_size)
+        layers = []
+        in_c = obs_shape[0]
+        for i, (k, s, p) in enumerate(zip(kernel_size, stride, padding)):
+            layers.append(
+                conv2d_block(in_c, hidden_size_list[i], k, s, p, activation=activation, norm_type=norm_type)
+            )
+            in_c = hidden_size_list[i]
+        layers.append(nn.Flatten())
+        self.main = nn.Sequential(*layers)
+        flatten_size = self._get_flatten_size()
+        self.output_size = hidden_size_list[-1]
+        self.mid = nn.Linear(flatten_size, hidden_size_list[-1])
+        self.act = build_activation(activation)
+
+    def _get_flatten_size(self) -> int:
+        with torch.no_grad():
+            test = torch.zeros(1, *self.obs_shape)
+            return self.main(test).shape[1]
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.act(self.mid(self.main(x)))
+
+
+class FCEncoder(nn.Module):
+    """MLP encoder for vector observations."""
+
+    def __init__(
+        self,
+        obs_shape: int,
+        hidden_size_list: List[int],
+        res_block: bool = False,
+        activation: str = 'relu',
+        norm_type: Optional[str] = None,
+        dropout: Optional[float] = None,
+    ):
+        super().__init__()
+        self.obs_shape = obs_shape
+        act = activation
+        self.init = nn.Linear(obs_shape, hidden_size_list[0])
+        self.act = build_activation(activation)
+        if res_block:
+            assert len(set(hidden_size_list)) == 1, "res_block requires constant width"
+            blocks = [ResFCBlock(hidden_size_list[0], activation=act, norm_type=norm_type, dropout=dropout)
+                      for _ in range(len(hidden_size_list))]
+            self.main = nn.Sequential(*blocks)
+        else:
+            layers = []
+            for i in range(len(hidden_size_list) - 1):
+                layers.append(
+                    fc_block(hidden_size_list[i], hidden_size_list[i + 1], activation=act, norm_type=norm_type,
+                             use_dropout=dropout is not None, dropout_probability=dropout or 0.5)
+                )
+            self.main = nn.Sequential(*layers)
+        self.output_size = hidden_size_list[-1]
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.main(self.act(self.init(x)))
+
+
+class StructEncoder(nn.Module):
+    """Dict-obs encoder: independent sub-encoders concatenated."""
+
+    def __init__(self, encoders: Dict[str, nn.Module]):
+        super().__init__()
+        self.encoders = nn.ModuleDict(encoders)
+        self.output_size = sum(getattr(e, 'output_size', 0) for e in encoders.values())
+
+    def forward(self, x: Dict[str, torch.Tensor]) -> torch.Tensor:
+        return torch.cat([self.encoders[k](x[k]) for k in sorted(self.encoders.keys())], dim=-1)
+
+
+class IMPALACnnResidualBlock(nn.Module):
+
+    def __init__(self, depth: int, scale: float = 1.0, batch_norm: bool = False):
+        super().__init__()
+        self.scale = scale
+        s = math.sqrt(scale)
+        self.conv0 = normed_conv2d(depth, depth, 3, padding=1, scale=s)
+        self.conv1 = normed_conv2d(depth, depth, 3, padding=1, scale=s)
+        self.bn0 = nn.BatchNorm2d(depth) if batch_norm else None
+        self.bn1 = nn.BatchNorm2d(depth) if batch_norm else None
+
+    def forward(self, x):
+        out = x
+        if self.bn0 is not None:
+            out = self.bn0(out)
+        out = self.conv0(torch.relu(out))
+        if self.bn1 is not None:
+            out = self.bn1(out)
+        out = self.conv1(torch.relu(out))
+        return x + out
+
+
+class IMPALACnnDownStack(nn.Module):
+    """conv -> maxpool -> nblock residual blocks."""
+
+    def __init__(self, in_c: int, nblock: int, out_c: int, scale: float = 1.0, pool: bool = True, **kwargs):
+        super().__init__()
+        self.pool = pool
+        self.firstconv = normed_conv2d(in_c, out_c, 3, padding=1)
+        s = scale / math.sqrt(nblock)
+        self.blocks = nn.ModuleList([IMPALACnnResidualBlock(out_c, scale=s, **kwargs) for _ in range(nblock)])
+
+    def forward(self, x):
+        x = self.firstconv(x)
+        if self.pool:
+            x = torch.nn.functional.max_pool2d(x, kernel_size=3, stride=2, padding=1)
+        for b in self.blocks:
+            x = b(x)
+        return x
+
+
+class IMPALAConvEncoder(nn.Module):
+    """IMPALA resnet encoder (dm-style): 3 down-stacks + fc."""
+
+    name = "IMPALAConvEncoder"
+
+    def __init__(
+        self,
+        obs_shape: tuple,
+        channels: tuple = (16, 32, 32),
+        outsize: int = 256,
+        scale_ob: float = 255.0,
+        nblock: int = 2,
+        final_relu: bool = True,
+        **kwargs,
+    ):
+        super().__init__()
+        self.scale_ob = scale_ob
+        c, h, w = obs_shape
+        curshape = (c, h, w)
+        s = 1 / math.sqrt(len(channels))
+        self.stacks = nn.ModuleList()
+        for out_c in channels:
+            stack = IMPALACnnDownStack(curshape[0], nblock=nblock, out_c=out_c, scale=s, **kwargs)
+            self.stacks.append(stack)
+            curshape = (out_c, (curshape[1] + 1) // 2, (curshape[2] + 1) // 2)
+        self.dense = normed_linear(prod(curshape), outsize, scale=1.4)
+        self.outsize = outsize
+        self.output_size = outsize
+        self.final_relu = final_relu
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        x = x.float() / self.scale_ob
+        for stack in self.stacks:
+            x = stack(x)
+        x = x.reshape(x.shape[0], -1)
+        x = torch.relu(x)
+        x = self.dense(x)
+        if self.final_relu:
+            x = torch.relu(x)
+        return x
+
+
+class GaussianFourierProjectionTimeEncoder(nn.Module):
+    """Random-Fourier time embedding (diffusion models)."""
+
+    def __init__(self, embed_dim: int, scale: float = 30.0):
+        super().__init__()
+        self.W = nn.Parameter(torch.randn(embed_dim // 2) * scale, requires_grad=False)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        x_proj = x[..., None] * self.W[None, :] * 2 * math.pi
+        return torch.cat([torch.sin(x_proj), torch.cos(x_proj)], dim=-1)

@@ -1,0 +1,3 @@
+from .q_learning import DQN, BDQ, C51DQN, QRDQN, IQN, FQF, RainbowDQN, DRQN, GTrXLDQN
+from .vac import VAC
+from .qac import ContinuousQAC, DiscreteQAC
