@@ -1,0 +1,16 @@
+from .env.base_env import BaseEnv, BaseEnvTimestep, get_vec_env_setting, get_env_cls, create_env
+from .env.ding_env_wrapper import DingEnvWrapper
+from .env.env_implementation_check import check_env_implementation
+from .env_wrappers.env_wrappers import (
+    EnvWrapper, NoopResetWrapper, MaxAndSkipWrapper, WarpFrameWrapper, ScaledFloatFrameWrapper, ClipRewardWrapper,
+    FrameStackWrapper, ObsNormWrapper, RewardNormWrapper, EpisodicLifeWrapper, FireResetWrapper, TimeLimitWrapper,
+    DelayRewardWrapper, EvalEpisodeReturnWrapper, ObsTransposeWrapper, FlatObsWrapper, ActionRepeatWrapper,
+    update_shape,
+)
+from .env_manager.base_env_manager import (
+    BaseEnvManager, BaseEnvManagerV2, EnvState, create_env_manager, get_env_manager_cls,
+)
+from .env_manager.subprocess_env_manager import (
+    SyncSubprocessEnvManager, AsyncSubprocessEnvManager, SubprocessEnvManagerV2,
+)
+from .common.spaces import Discrete, Box, MultiDiscrete

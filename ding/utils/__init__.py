@@ -26,6 +26,7 @@ from .dist_helper import (
     simple_group_split, dist_init, dist_finalize, DistContext, DDPContext, synchronize,
     is_dist_initialized,
 )
+from .import_helper import import_module, try_import_ceph, try_import_redis
 from .scheduler_helper import Scheduler
 from .normalizer_helper import DatasetNormalizer
 from .fast_copy import fast_copy
