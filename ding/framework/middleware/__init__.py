@@ -1,0 +1,6 @@
+from .functional import *  # noqa
+from .collector import StepCollector, EpisodeCollector
+from .learner import OffPolicyLearner, HERLearner
+from .ckpt_handler import CkptSaver
+from .distributer import ContextExchanger, ModelExchanger, PeriodicalModelExchanger
+from .barrier import Barrier, BarrierRuntime

@@ -34,6 +34,13 @@ class IModelWrapper(nn.Module):
     def model(self):
         return self._model
 
+    def state_dict(self, *args, **kwargs):
+        # delegate to the innermost raw model so checkpoints are wrapper-free
+        return self._model.state_dict(*args, **kwargs)
+
+    def load_state_dict(self, state_dict, *args, **kwargs):
+        return self._model.load_state_dict(state_dict, *args, **kwargs)
+
     def forward(self, *args, **kwargs):
         return self._model.forward(*args, **kwargs)
 

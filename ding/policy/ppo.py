@@ -1,0 +1,484 @@
+"""PPO policies: on-policy PPO (discrete/continuous/hybrid, multi-agent
+aware), PPO-PG (no critic), off-policy PPO.
+
+Parity: reference ding/policy/ppo.py ('ppo', 'ppo_pg', 'ppo_offpolicy',
+'ppo_stdim' registrations, ~1,900 LoC).
+"""
+import copy
+from collections import namedtuple
+from typing import Any, Dict, List, Optional
+
+import numpy as np
+import torch
+
+from ding.model import model_wrap
+from ding.rl_utils import (
+    ppo_data, ppo_error, ppo_error_continuous, ppo_policy_data, ppo_policy_error, gae, gae_data, get_gae,
+    get_train_sample, ppo_policy_error_continuous, get_nstep_return_data,
+)
+from ding.torch_utils import Adam, to_device
+from ding.utils import split_data_generator
+from ding.utils import POLICY_REGISTRY
+from ding.utils.data import default_collate, default_decollate
+from .base_policy import Policy
+from .common_utils import default_preprocess_learn
+
+
+@POLICY_REGISTRY.register('ppo')
+class PPOPolicy(Policy):
+    """On-policy PPO with GAE, value/dual clip, recompute-adv epochs."""
+
+    config = dict(
+        type='ppo',
+        cuda=False,
+        on_policy=True,
+        priority=False,
+        priority_IS_weight=False,
+        recompute_adv=True,
+        action_space='discrete',
+        nstep_return=False,
+        multi_agent=False,
+        transition_with_policy_data=True,
+        model=dict(),
+        learn=dict(
+            epoch_per_collect=10,
+            batch_size=64,
+            learning_rate=3e-4,
+            value_weight=0.5,
+            entropy_weight=0.01,
+            clip_ratio=0.2,
+            adv_norm=True,
+            value_norm=True,
+            ppo_param_init=True,
+            grad_clip_type='clip_norm',
+            grad_clip_value=0.5,
+            ignore_done=False,
+        ),
+        collect=dict(
+            unroll_len=1,
+            discount_factor=0.99,
+            gae_lambda=0.95,
+        ),
+        eval=dict(),
+    )
+
+    def default_model(self) -> tuple:
+        if self._cfg.multi_agent:
+            return 'mavac', ['ding.model.template.mavac']
+        return 'vac', ['ding.model.template.vac']
+
+    def _init_learn(self) -> None:
+        self._action_space = self._cfg.action_space
+        assert self._action_space in ('discrete', 'continuous', 'hybrid')
+        if self._cfg.learn.ppo_param_init:
+            for n, m in self._model.named_modules():
+                if isinstance(m, torch.nn.Linear):
+                    torch.nn.init.orthogonal_(m.weight, gain=np.sqrt(2))
+                    torch.nn.init.zeros_(m.bias)
+            if self._action_space in ('continuous', 'hybrid'):
+                for m in getattr(self._model.actor_head, 'modules', lambda: [])():
+                    if isinstance(m, torch.nn.Linear):
+                        torch.nn.init.zeros_(m.bias)
+                        m.weight.data.copy_(0.01 * m.weight.data)
+
+        self._optimizer = Adam(
+            self._model.parameters(),
+            lr=self._cfg.learn.learning_rate,
+            grad_clip_type=self._cfg.learn.grad_clip_type,
+            clip_value=self._cfg.learn.grad_clip_value,
+        )
+        self._learn_model = model_wrap(self._model, wrapper_name='base')
+        self._value_weight = self._cfg.learn.value_weight
+        self._entropy_weight = self._cfg.learn.entropy_weight
+        self._clip_ratio = self._cfg.learn.clip_ratio
+        self._adv_norm = self._cfg.learn.adv_norm
+        self._value_norm = self._cfg.learn.value_norm
+        if self._value_norm:
+            from ding.utils import RunningMeanStd
+            self._running_mean_std = RunningMeanStd(epsilon=1e-4, shape=(1, ))
+        self._gamma = self._cfg.collect.discount_factor
+        self._gae_lambda = self._cfg.collect.gae_lambda
+        self._recompute_adv = self._cfg.recompute_adv
+        self._learn_model.reset()
+
+    def _forward_learn(self, data: List[Dict[str, Any]]) -> List[Dict[str, Any]]:
+        data = default_preprocess_learn(data, ignore_done=self._cfg.learn.ignore_done, use_nstep=False)
+        if self._cuda:
+            data = to_device(data, self._device)
+        data['obs'] = data['obs'].float()
+        if 'next_obs' in data:
+            data['next_obs'] = data['next_obs'].float()
+        self._learn_model.train()
+        return_infos = []
+        for epoch in range(self._cfg.learn.epoch_per_collect):
+            if self._recompute_adv:
+                with torch.no_grad():
+                    value = self._learn_model.forward(data['obs'], mode='compute_critic')['value']
+                    next_value = self._learn_model.forward(data['next_obs'], mode='compute_critic')['value']
+                    if self._value_norm:
+                        value *= float(self._running_mean_std.std[0])
+                        next_value *= float(self._running_mean_std.std[0])
+                    traj_flag = data.get('traj_flag')
+                    compute_adv_data = gae_data(value, next_value, data['reward'], data['done'], traj_flag)
+                    data['adv'] = gae(compute_adv_data, self._gamma, self._gae_lambda)
+                    unnormalized_returns = value + data['adv']
+                    if self._value_norm:
+                        data['value'] = value / float(self._running_mean_std.std[0])
+                        data['return'] = unnormalized_returns / float(self._running_mean_std.std[0])
+                        self._running_mean_std.update(unnormalized_returns.cpu().numpy().reshape(-1, 1))
+                    else:
+                        data['value'] = value
+                        data['return'] = unnormalized_returns
+            else:
+                if self._value_norm:
+                    unnormalized_return = data['adv'] + data['value'] * float(self._running_mean_std.std[0])
+                    data['return'] = unnormalized_return / float(self._running_mean_std.std[0])
+                    self._running_mean_std.update(unnormalized_return.cpu().numpy().reshape(-1, 1))
+                else:
+                    data['return'] = data['adv'] + data['value']
+
+            for batch in split_data_generator(data, self._cfg.learn.batch_size, shuffle=True):
+                output = self._learn_model.forward(batch['obs'], mode='compute_actor_critic')
+                adv = batch['adv']
+                if self._adv_norm:
+                    adv = (adv - adv.mean()) / (adv.std() + 1e-8)
+                if self._action_space == 'continuous':
+                    ppo_batch = ppo_data(
+                        output['logit'], batch['logit'], batch['action'], output['value'], batch['value'], adv,
+                        batch['return'], batch.get('weight')
+                    )
+                    ppo_loss, ppo_info = ppo_error_continuous(ppo_batch, self._clip_ratio)
+                elif self._action_space == 'discrete':
+                    ppo_batch = ppo_data(
+                        output['logit'], batch['logit'], batch['action'], output['value'], batch['value'], adv,
+                        batch['return'], batch.get('weight')
+                    )
+                    ppo_loss, ppo_info = ppo_error(ppo_batch, self._clip_ratio)
+                else:  # hybrid
+                    # action-type part
+                    type_batch = ppo_policy_data(
+                        output['logit']['action_type'], batch['logit']['action_type'], batch['action']['action_type'],
+                        adv, batch.get('weight')
+                    )
+                    type_loss, type_info = ppo_policy_error(type_batch, self._clip_ratio)
+                    from ding.rl_utils.ppo import ppo_policy_data_continuous, ppo_value_data, ppo_value_error
+                    args_batch = ppo_policy_data_continuous(
+                        output['logit']['action_args'], batch['logit']['action_args'],
+                        batch['action']['action_args'], adv, batch.get('weight')
+                    )
+                    args_loss, args_info = ppo_policy_error_continuous(args_batch, self._clip_ratio)
+                    value_loss = ppo_value_error(
+                        ppo_value_data(output['value'], batch['value'], batch['return'], batch.get('weight')),
+                        self._clip_ratio
+                    )
+                    from ding.rl_utils.ppo import ppo_loss as ppo_loss_tuple, ppo_info as ppo_info_tuple
+                    ppo_loss = ppo_loss_tuple(
+                        type_loss.policy_loss + args_loss.policy_loss, value_loss,
+                        type_loss.entropy_loss + args_loss.entropy_loss, torch.zeros(())
+                    )
+                    ppo_info = ppo_info_tuple(
+                        max(type_info.approx_kl, args_info.approx_kl), max(type_info.clipfrac, args_info.clipfrac)
+                    )
+                wv, we = self._value_weight, self._entropy_weight
+                total_loss = ppo_loss.policy_loss + wv * ppo_loss.value_loss - we * ppo_loss.entropy_loss
+                self._optimizer.zero_grad()
+                total_loss.backward()
+                if self._cfg.multi_gpu:
+                    self.sync_gradients(self._model)
+                self._optimizer.step()
+
+                return_info = {
+                    'cur_lr': self._optimizer.defaults['lr'],
+                    'total_loss': total_loss.item(),
+                    'policy_loss': ppo_loss.policy_loss.item(),
+                    'value_loss': ppo_loss.value_loss.item(),
+                    'entropy_loss': ppo_loss.entropy_loss.item(),
+                    'adv_max': adv.max().item(),
+                    'adv_mean': adv.mean().item(),
+                    'value_mean': output['value'].mean().item(),
+                    'value_max': output['value'].max().item(),
+                    'approx_kl': ppo_info.approx_kl,
+                    'clipfrac': ppo_info.clipfrac,
+                }
+                return_infos.append(return_info)
+        return return_infos
+
+    def _monitor_vars_learn(self) -> List[str]:
+        variables = [
+            'cur_lr', 'total_loss', 'policy_loss', 'value_loss', 'entropy_loss', 'adv_max', 'adv_mean',
+            'approx_kl', 'clipfrac', 'value_max', 'value_mean',
+        ]
+        return variables
+
+    def _init_collect(self) -> None:
+        self._unroll_len = self._cfg.collect.unroll_len
+        self._action_space = self._cfg.action_space
+        if self._action_space == 'continuous':
+            self._collect_model = model_wrap(self._model, wrapper_name='reparam_sample')
+        elif self._action_space == 'discrete':
+            self._collect_model = model_wrap(self._model, wrapper_name='multinomial_sample')
+        else:
+            self._collect_model = model_wrap(self._model, wrapper_name='hybrid_reparam_multinomial_sample')
+        self._collect_model.reset()
+        self._gamma = self._cfg.collect.discount_factor
+        self._gae_lambda = self._cfg.collect.gae_lambda
+        self._recompute_adv = self._cfg.recompute_adv
+
+    def _forward_collect(self, data: Dict[int, Any], **kwargs) -> Dict[int, Any]:
+        data_id = list(data.keys())
+        data = default_collate(list(data.values()))
+        if self._cuda:
+            data = to_device(data, self._device)
+        self._collect_model.eval()
+        with torch.no_grad():
+            output = self._collect_model.forward(data, mode='compute_actor_critic')
+        if self._cuda:
+            output = to_device(output, 'cpu')
+        output = default_decollate(output)
+        return {i: d for i, d in zip(data_id, output)}
+
+    def _process_transition(self, obs: Any, policy_output: Dict[str, Any], timestep: namedtuple) -> Dict[str, Any]:
+        return {
+            'obs': obs,
+            'next_obs': timestep.obs,
+            'action': policy_output['action'],
+            'logit': policy_output['logit'],
+            'value': policy_output['value'],
+            'reward': timestep.reward,
+            'done': timestep.done,
+        }
+
+    def _get_train_sample(self, transitions: List[Dict[str, Any]]) -> List[Dict[str, Any]]:
+        data = transitions
+        data = get_gae_with_default_last_value_wrapper(
+            data, done=data[-1]['done'], gamma=self._gamma, gae_lambda=self._gae_lambda, cuda=False
+        )
+        return get_train_sample(data, self._unroll_len)
+
+    def _init_eval(self) -> None:
+        self._action_space = self._cfg.action_space
+        if self._action_space == 'continuous':
+            self._eval_model = model_wrap(self._model, wrapper_name='deterministic_sample')
+        elif self._action_space == 'discrete':
+            self._eval_model = model_wrap(self._model, wrapper_name='argmax_sample')
+        else:
+            self._eval_model = model_wrap(self._model, wrapper_name='hybrid_deterministic_argmax_sample')
+        self._eval_model.reset()
+
+    def _forward_eval(self, data: Dict[int, Any]) -> Dict[int, Any]:
+        data_id = list(data.keys())
+        data = default_collate(list(data.values()))
+        if self._cuda:
+            data = to_device(data, self._device)
+        self._eval_model.eval()
+        with torch.no_grad():
+            output = self._eval_model.forward(data, mode='compute_actor')
+        if self._cuda:
+            output = to_device(output, 'cpu')
+        output = default_decollate(output)
+        return {i: d for i, d in zip(data_id, output)}
+
+
+def get_gae_with_default_last_value_wrapper(data, done, gamma, gae_lambda, cuda):
+    from collections import deque
+    from ding.rl_utils import Adder
+    return Adder.get_gae_with_default_last_value(deque(data), done, gamma, gae_lambda, cuda)
+
+
+@POLICY_REGISTRY.register('ppo_pg')
+class PPOPGPolicy(Policy):
+    """PPO policy-gradient-only variant (no value function; MC returns)."""
+
+    config = dict(
+        type='ppo_pg',
+        cuda=False,
+        on_policy=True,
+        action_space='discrete',
+        model=dict(),
+        learn=dict(
+            epoch_per_collect=10,
+            batch_size=64,
+            learning_rate=3e-4,
+            entropy_weight=0.01,
+            clip_ratio=0.2,
+            grad_clip_type='clip_norm',
+            grad_clip_value=0.5,
+            ignore_done=False,
+        ),
+        collect=dict(
+            unroll_len=1,
+            discount_factor=0.99,
+        ),
+        eval=dict(),
+    )
+
+    def default_model(self) -> tuple:
+        return 'pg', ['ding.model.template.pg']
+
+    def _init_learn(self) -> None:
+        self._optimizer = Adam(
+            self._model.parameters(), lr=self._cfg.learn.learning_rate,
+            grad_clip_type=self._cfg.learn.grad_clip_type, clip_value=self._cfg.learn.grad_clip_value
+        )
+        self._learn_model = model_wrap(self._model, wrapper_name='base')
+        self._entropy_weight = self._cfg.learn.entropy_weight
+        self._clip_ratio = self._cfg.learn.clip_ratio
+        self._action_space = self._cfg.action_space
+        self._learn_model.reset()
+
+    def _forward_learn(self, data: List[Dict[str, Any]]) -> List[Dict[str, Any]]:
+        data = default_preprocess_learn(data)
+        if self._cuda:
+            data = to_device(data, self._device)
+        self._learn_model.train()
+        return_infos = []
+        for epoch in range(self._cfg.learn.epoch_per_collect):
+            for batch in split_data_generator(data, self._cfg.learn.batch_size, shuffle=True):
+                output = self._learn_model.forward(batch['obs'])
+                adv = batch['return']
+                adv = (adv - adv.mean()) / (adv.std() + 1e-8)
+                if self._action_space == 'discrete':
+                    pol_data = ppo_policy_data(output['logit'], batch['logit'], batch['action'], adv,
+                                               batch.get('weight'))
+                    loss_t, info = ppo_policy_error(pol_data, self._clip_ratio)
+                else:
+                    from ding.rl_utils.ppo import ppo_policy_data_continuous
+                    pol_data = ppo_policy_data_continuous(output['logit'], batch['logit'], batch['action'], adv,
+                                                          batch.get('weight'))
+                    loss_t, info = ppo_policy_error_continuous(pol_data, self._clip_ratio)
+                total_loss = loss_t.policy_loss - self._entropy_weight * loss_t.entropy_loss
+                self._optimizer.zero_grad()
+                total_loss.backward()
+                if self._cfg.multi_gpu:
+                    self.sync_gradients(self._model)
+                self._optimizer.step()
+                return_infos.append({
+                    'cur_lr': self._optimizer.defaults['lr'],
+                    'total_loss': total_loss.item(),
+                    'policy_loss': loss_t.policy_loss.item(),
+                    'entropy_loss': loss_t.entropy_loss.item(),
+                    'approx_kl': info.approx_kl,
+                    'clipfrac': info.clipfrac,
+                })
+        return return_infos
+
+    def _monitor_vars_learn(self) -> List[str]:
+        return ['cur_lr', 'total_loss', 'policy_loss', 'entropy_loss', 'approx_kl', 'clipfrac']
+
+    def _init_collect(self) -> None:
+        self._unroll_len = self._cfg.collect.unroll_len
+        self._gamma = self._cfg.collect.discount_factor
+        if self._cfg.action_space == 'discrete':
+            self._collect_model = model_wrap(self._model, wrapper_name='multinomial_sample')
+        else:
+            self._collect_model = model_wrap(self._model, wrapper_name='reparam_sample')
+        self._collect_model.reset()
+
+    def _forward_collect(self, data: Dict[int, Any], **kwargs) -> Dict[int, Any]:
+        data_id = list(data.keys())
+        data = default_collate(list(data.values()))
+        if self._cuda:
+            data = to_device(data, self._device)
+        self._collect_model.eval()
+        with torch.no_grad():
+            output = self._collect_model.forward(data)
+        if self._cuda:
+            output = to_device(output, 'cpu')
+        output = default_decollate(output)
+        return {i: d for i, d in zip(data_id, output)}
+
+    def _process_transition(self, obs, policy_output, timestep) -> Dict[str, Any]:
+        return {
+            'obs': obs,
+            'action': policy_output['action'],
+            'logit': policy_output['logit'],
+            'reward': timestep.reward,
+            'done': timestep.done,
+        }
+
+    def _get_train_sample(self, data: List[Dict[str, Any]]) -> List[Dict[str, Any]]:
+        # Monte-Carlo returns over the episode
+        R = 0.0
+        for i in reversed(range(len(data))):
+            R = self._gamma * R + data[i]['reward'].item()
+            data[i]['return'] = torch.tensor([R])
+        return get_train_sample(data, self._unroll_len)
+
+    def _init_eval(self) -> None:
+        if self._cfg.action_space == 'discrete':
+            self._eval_model = model_wrap(self._model, wrapper_name='argmax_sample')
+        else:
+            self._eval_model = model_wrap(self._model, wrapper_name='deterministic_sample')
+        self._eval_model.reset()
+
+    def _forward_eval(self, data: Dict[int, Any]) -> Dict[int, Any]:
+        data_id = list(data.keys())
+        data = default_collate(list(data.values()))
+        if self._cuda:
+            data = to_device(data, self._device)
+        self._eval_model.eval()
+        with torch.no_grad():
+            output = self._eval_model.forward(data)
+        if self._cuda:
+            output = to_device(output, 'cpu')
+        output = default_decollate(output)
+        return {i: d for i, d in zip(data_id, output)}
+
+
+@POLICY_REGISTRY.register('ppo_offpolicy')
+class PPOOffPolicy(PPOPolicy):
+    """Off-policy PPO: GAE computed at collect time, samples pushed to a
+    replay buffer, nstep-return option."""
+
+    config = dict(
+        type='ppo_offpolicy',
+        on_policy=False,
+        recompute_adv=False,
+        learn=dict(
+            update_per_collect=5,
+            batch_size=64,
+        ),
+        other=dict(replay_buffer=dict(replay_buffer_size=10000, )),
+    )
+
+    def _forward_learn(self, data: List[Dict[str, Any]]) -> Dict[str, Any]:
+        data = default_preprocess_learn(data, ignore_done=self._cfg.learn.ignore_done, use_nstep=False)
+        if self._cuda:
+            data = to_device(data, self._device)
+        self._learn_model.train()
+        data['obs'] = data['obs'].float()
+        data['return'] = data['adv'] + data['value']
+        output = self._learn_model.forward(data['obs'], mode='compute_actor_critic')
+        adv = data['adv']
+        if self._adv_norm:
+            adv = (adv - adv.mean()) / (adv.std() + 1e-8)
+        ppo_batch = ppo_data(
+            output['logit'], data['logit'], data['action'], output['value'], data['value'], adv, data['return'],
+            data.get('weight')
+        )
+        if self._action_space == 'continuous':
+            ppo_loss, ppo_info = ppo_error_continuous(ppo_batch, self._clip_ratio)
+        else:
+            ppo_loss, ppo_info = ppo_error(ppo_batch, self._clip_ratio)
+        total_loss = ppo_loss.policy_loss + self._value_weight * ppo_loss.value_loss \
+            - self._entropy_weight * ppo_loss.entropy_loss
+        self._optimizer.zero_grad()
+        total_loss.backward()
+        if self._cfg.multi_gpu:
+            self.sync_gradients(self._model)
+        self._optimizer.step()
+        return {
+            'cur_lr': self._optimizer.defaults['lr'],
+            'total_loss': total_loss.item(),
+            'policy_loss': ppo_loss.policy_loss.item(),
+            'value_loss': ppo_loss.value_loss.item(),
+            'entropy_loss': ppo_loss.entropy_loss.item(),
+            'approx_kl': ppo_info.approx_kl,
+            'clipfrac': ppo_info.clipfrac,
+        }
+
+    def _get_train_sample(self, transitions: List[Dict[str, Any]]) -> List[Dict[str, Any]]:
+        data = get_gae_with_default_last_value_wrapper(
+            transitions, done=transitions[-1]['done'], gamma=self._gamma, gae_lambda=self._gae_lambda, cuda=False
+        )
+        return get_train_sample(data, self._unroll_len)
