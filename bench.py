@@ -1,0 +1,232 @@
+"""Flagship benchmark: Atari PPO learner throughput on MI355X.
+
+Measures the BASELINE.json headline metric — learner samples/sec
+(env-steps/sec) for the Atari Pong PPO workload (reference config
+dizoo/atari/config/serial/pong/pong_ppo_config.py: n_sample=3200,
+batch=320, epoch_per_collect=10, conv encoder [64,64,128] on 4x84x84
+frames) — on synthetic env transitions and random-init weights (no network
+for datasets). One bench step = one full PPO train phase over a freshly
+generated 3200-sample batch: 10 epochs x 10 minibatches of 320, with
+advantage recomputation each epoch, forward+backward+optimizer step all
+inside the timed region. fp32 compute (same precision as the reference's
+training).
+
+Also supports the IMPALA workload (--workload impala): unroll_len=32,
+batch=128 trajectories, v-trace loss (reference
+spaceinvaders_impala_config.py:17-45).
+
+Multi-GPU: launched by the driver via torch.distributed.run with one rank
+per GPU over RCCL; weak scaling (each rank trains its own 3200-sample
+batch; gradients all-reduced through the bucketed reducer).
+"""
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def build_ppo_policy(device: str, multi_gpu: bool):
+    from ding.policy import PPOPolicy
+    from ding.utils import EasyDict, deep_merge_dicts
+
+    cfg = PPOPolicy.default_config()
+    user = dict(
+        cuda=device.startswith("cuda"),
+        multi_gpu=multi_gpu,
+        action_space='discrete',
+        recompute_adv=True,
+        model=dict(
+            obs_shape=[4, 84, 84],
+            action_shape=6,
+            action_space='discrete',
+            encoder_hidden_size_list=[64, 64, 128],
+            actor_head_hidden_size=128,
+            critic_head_hidden_size=128,
+        ),
+        learn=dict(
+            epoch_per_collect=10,
+            batch_size=320,
+            learning_rate=3e-4,
+            value_weight=0.5,
+            entropy_weight=0.001,
+            clip_ratio=0.2,
+            adv_norm=True,
+            value_norm=True,
+            ignore_done=False,
+            grad_clip_type='clip_norm',
+            grad_clip_value=0.5,
+        ),
+        collect=dict(n_sample=3200, unroll_len=1, discount_factor=0.99, gae_lambda=0.95),
+    )
+    cfg = EasyDict(deep_merge_dicts(cfg, user))
+    return PPOPolicy(cfg, enable_field=['learn'])
+
+
+def make_ppo_batch(n_sample: int, device: str, policy) -> dict:
+    """Synthetic Pong-shaped transitions resident on the GPU."""
+    obs = torch.randint(0, 255, (n_sample, 4, 84, 84), dtype=torch.uint8, device=device)
+    next_obs = torch.randint(0, 255, (n_sample, 4, 84, 84), dtype=torch.uint8, device=device)
+    with torch.no_grad():
+        # behaviour logits/values from the current net (realistic distributions)
+        out = policy._model(obs[:320].float().div_(255.0), mode='compute_actor_critic')
+    B = n_sample
+    logit = out['logit'].detach().repeat((B + 319) // 320, 1)[:B].contiguous()
+    value = out['value'].detach().repeat((B + 319) // 320)[:B].contiguous()
+    action = torch.distributions.Categorical(logits=logit).sample()
+    reward = torch.sign(torch.randn(B, device=device)) * (torch.rand(B, device=device) < 0.05)
+    done = (torch.rand(B, device=device) < 0.002).float()
+    return {
+        'obs': obs,
+        'next_obs': next_obs,
+        'action': action,
+        'logit': logit,
+        'value': value,
+        'adv': torch.randn(B, device=device),
+        'reward': reward,
+        'done': done,
+        'weight': None,
+    }
+
+
+def ppo_step(policy, device: str, n_sample: int):
+    data = make_ppo_batch(n_sample, device, policy)
+    # scale obs like the Atari pipeline (uint8 -> [0,1]) without a CPU trip
+    data['obs'] = data['obs'].float().div_(255.0)
+    data['next_obs'] = data['next_obs'].float().div_(255.0)
+    policy._forward_learn(data)
+    return n_sample
+
+
+def build_impala(device: str, multi_gpu: bool):
+    """IMPALA workload: model + optimizer + v-trace loss (policy class lands
+    in a later milestone; the learner math path is exercised directly)."""
+    from ding.model import VAC
+    from ding.torch_utils import Adam
+    model = VAC(
+        obs_shape=[4, 84, 84], action_shape=6, encoder_hidden_size_list=[128, 128, 256],
+        actor_head_hidden_size=256, critic_head_hidden_size=256
+    ).to(device)
+    optimizer = Adam(model.parameters(), lr=6e-4, grad_clip_type='clip_norm', clip_value=5)
+    return model, optimizer
+
+
+def impala_step(model, optimizer, device: str, batch_size: int = 128, unroll_len: int = 32):
+    from ding.rl_utils import vtrace_data, vtrace_error_discrete_action
+    T, B = unroll_len, batch_size
+    obs = torch.rand(T + 1, B, 4, 84, 84, device=device)
+    flat = obs.view((T + 1) * B, 4, 84, 84)
+    out = model(flat, mode='compute_actor_critic')
+    logit = out['logit'].view(T + 1, B, -1)
+    value = out['value'].view(T + 1, B)
+    action = torch.randint(0, 6, (T, B), device=device)
+    reward = torch.randn(T, B, device=device)
+    behaviour = logit[:-1].detach() + 0.1 * torch.randn_like(logit[:-1])
+    data = vtrace_data(logit[:-1], behaviour, action, value, reward, None)
+    loss = vtrace_error_discrete_action(data, gamma=0.99, lambda_=0.95)
+    total = loss.policy_loss + 0.5 * loss.value_loss - 0.01 * loss.entropy_loss
+    optimizer.zero_grad()
+    total.backward()
+    if torch.distributed.is_initialized():
+        from ding.parallel import sync_gradients_flat
+        sync_gradients_flat(model)
+    optimizer.step()
+    return T * B
+
+
+def main():
+    parser = argparse.ArgumentParser()
+    parser.add_argument('--gpus', type=int, default=1)
+    parser.add_argument('--steps', type=int, default=5)
+    parser.add_argument('--warmup', type=int, default=2)
+    parser.add_argument('--workload', type=str, default='ppo', choices=['ppo', 'impala'])
+    parser.add_argument('--n-sample', type=int, default=3200)
+    args = parser.parse_args()
+
+    world_size = int(os.environ.get('WORLD_SIZE', '1'))
+    rank = int(os.environ.get('RANK', '0'))
+    local_rank = int(os.environ.get('LOCAL_RANK', rank))
+    distributed = world_size > 1
+
+    use_gpu = torch.cuda.is_available()
+    if use_gpu:
+        torch.cuda.set_device(local_rank % max(1, torch.cuda.device_count()))
+        device = f'cuda:{torch.cuda.current_device()}'
+    else:
+        device = 'cpu'
+
+    if distributed:
+        os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
+        os.environ.setdefault('MASTER_PORT', '29500')
+        torch.distributed.init_process_group(backend='nccl' if use_gpu else 'gloo')
+
+    torch.manual_seed(1234 + rank)
+
+    if args.workload == 'ppo':
+        policy = build_ppo_policy(device, multi_gpu=distributed)
+        step_fn = lambda: ppo_step(policy, device, args.n_sample)
+        model_name = 'pong_ppo(conv[64,64,128] 4x84x84)'
+        config = {
+            'model': model_name, 'global_batch': args.n_sample * world_size, 'seq_len': 1,
+            'parallelism': f'dp{world_size}', 'minibatch': 320, 'epoch_per_collect': 10,
+        }
+    else:
+        model, optimizer = build_impala(device, multi_gpu=distributed)
+        step_fn = lambda: impala_step(model, optimizer, device)
+        config = {
+            'model': 'spaceinvaders_impala(conv[128,128,256] 4x84x84)', 'global_batch': 128 * 32 * world_size,
+            'seq_len': 32, 'parallelism': f'dp{world_size}',
+        }
+
+    def barrier_sync():
+        if distributed:
+            torch.distributed.barrier()
+        if use_gpu:
+            torch.cuda.synchronize()
+
+    # warmup
+    for _ in range(args.warmup):
+        step_fn()
+    barrier_sync()
+
+    t0 = time.perf_counter()
+    samples = 0
+    for _ in range(args.steps):
+        samples += step_fn()
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if distributed:
+        t = torch.tensor([elapsed], device=device if use_gpu else 'cpu')
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = t.item()
+
+    total_samples = samples * world_size
+    value = total_samples / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        print(json.dumps({
+            'metric': 'learner samples/sec (env-steps/sec), Atari ' + args.workload.upper(),
+            'value': value,
+            'unit': 'samples/s',
+            'n_gpus': world_size,
+            'steps': args.steps,
+            'warmup': args.warmup,
+            'ms_per_step': ms_per_step,
+            'higher_is_better': True,
+            'scaling': 'weak',
+            'vs_baseline': None,
+            'dtype': 'fp32',
+            'data': 'synthetic',
+            'config': config,
+        }))
+
+    if distributed:
+        torch.distributed.destroy_process_group()
+
+
+if __name__ == '__main__':
+    main()

@@ -19,8 +19,12 @@ def default_preprocess_learn(
     ignore_done: bool = False,
 ) -> Dict[str, torch.Tensor]:
     """Collate a list of transition dicts into a train batch; normalize
-    reward layout for n-step ([B, T] -> [T, B]) and attach IS weights."""
-    data = default_collate(data)
+    reward layout for n-step ([B, T] -> [T, B]) and attach IS weights.
+
+    A dict input is treated as already collated (the MI355X middleware keeps
+    batches resident as stacked tensors instead of re-collating lists)."""
+    if not isinstance(data, dict):
+        data = default_collate(data)
     if 'value_gamma' in data and isinstance(data['value_gamma'], list):
         data['value_gamma'] = torch.as_tensor(data['value_gamma'], dtype=torch.float32)
     if ignore_done:

@@ -1,0 +1,160 @@
+"""HIP kernel numerics vs the pure-PyTorch fp32 oracle (DI-hpc harness
+pattern: mean relative error at the reference shapes, e.g. GAE T=1024 B=64).
+
+All tests require an MI355X (marked gpu)."""
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def mean_rel_err(a: torch.Tensor, b: torch.Tensor) -> float:
+    return ((a - b).abs() / (b.abs() + 1e-6)).mean().item()
+
+
+@pytest.fixture(scope="module")
+def ext():
+    from ding.ops import dispatch
+    assert dispatch.is_available(), "HIP extension must be built on the GPU box"
+    return dispatch
+
+
+def test_gae_hip_vs_oracle(ext):
+    from ding.rl_utils import gae, gae_data
+    T, B = 1024, 64
+    torch.manual_seed(0)
+    value = torch.randn(T, B, device="cuda")
+    next_value = torch.randn(T, B, device="cuda")
+    reward = torch.randn(T, B, device="cuda")
+    done = (torch.rand(T, B, device="cuda") < 0.01).float()
+    with torch.no_grad():
+        adv_hip = gae(gae_data(value, next_value, reward, done, None), 0.99, 0.95)
+    import os
+    os.environ["DI_ENGINE_DISABLE_HIP"] = "1"
+    try:
+        import importlib
+        import ding.ops.dispatch as d
+        importlib.reload(d)
+        with torch.no_grad():
+            adv_ref = gae(gae_data(value.clone(), next_value.clone(), reward, done, None), 0.99, 0.95)
+    finally:
+        os.environ.pop("DI_ENGINE_DISABLE_HIP")
+        import importlib
+        import ding.ops.dispatch as d
+        importlib.reload(d)
+    assert mean_rel_err(adv_hip, adv_ref) < 1e-4
+
+
+def test_reverse_scan_numerics(ext):
+    T, B = 1024, 64
+    delta = torch.randn(T, B, device="cuda")
+    factor = torch.rand(T, B, device="cuda") * 0.99
+    out = ext.gae_scan(delta, factor)
+    ref = torch.zeros_like(delta)
+    acc = torch.zeros(B, device="cuda")
+    for t in range(T - 1, -1, -1):
+        acc = delta[t] + factor[t] * acc
+        ref[t] = acc
+    assert torch.allclose(out, ref, atol=1e-3)
+
+
+def test_multistep_forward_view_hip(ext):
+    from ding.rl_utils import generalized_lambda_returns
+    T, B = 256, 128
+    value = torch.randn(T + 1, B, device="cuda")
+    reward = torch.randn(T, B, device="cuda")
+    with torch.no_grad():
+        out_hip = generalized_lambda_returns(value, reward, 0.9, 0.8)
+    # CPU oracle
+    with torch.no_grad():
+        out_ref = generalized_lambda_returns(value.cpu(), reward.cpu(), 0.9, 0.8)
+    assert mean_rel_err(out_hip.cpu(), out_ref) < 1e-4
+
+
+def test_td_lambda_and_vtrace_consistency(ext):
+    from ding.rl_utils import td_lambda_error, td_lambda_data, vtrace_error_discrete_action, vtrace_data
+    T, B, N = 64, 32, 8
+    value = torch.randn(T + 1, B, device="cuda", requires_grad=True)
+    reward = torch.rand(T, B, device="cuda")
+    loss_gpu = td_lambda_error(td_lambda_data(value, reward, None))
+    value_cpu = value.detach().cpu().requires_grad_(True)
+    loss_cpu = td_lambda_error(td_lambda_data(value_cpu, reward.cpu(), None))
+    assert abs(loss_gpu.item() - loss_cpu.item()) < 1e-3
+    # vtrace end to end on GPU
+    target = torch.randn(T, B, N, device="cuda", requires_grad=True)
+    behaviour = torch.randn(T, B, N, device="cuda")
+    action = torch.randint(0, N, (T, B), device="cuda")
+    loss = vtrace_error_discrete_action(vtrace_data(target, behaviour, action, value.detach(), reward, None))
+    total = loss.policy_loss + loss.value_loss
+    total.backward()
+    assert target.grad is not None
+
+
+def test_c51_project_hip_vs_oracle(ext):
+    from ding.rl_utils import dist_nstep_td_error, dist_nstep_td_data
+    B, N, A, nstep = 128, 6, 51, 3
+    torch.manual_seed(0)
+    dist = torch.softmax(torch.randn(B, N, A, device="cuda"), -1)
+    next_dist = torch.softmax(torch.randn(B, N, A, device="cuda"), -1)
+    act = torch.randint(0, N, (B, ), device="cuda")
+    next_act = torch.randint(0, N, (B, ), device="cuda")
+    reward = torch.randn(nstep, B, device="cuda")
+    done = (torch.rand(B, device="cuda") < 0.1).float()
+    data = dist_nstep_td_data(dist, next_dist, act, next_act, reward, done, None)
+    with torch.no_grad():
+        loss_hip, td_hip = dist_nstep_td_error(data, 0.99, -10, 10, A, nstep)
+    data_cpu = dist_nstep_td_data(dist.cpu(), next_dist.cpu(), act.cpu(), next_act.cpu(), reward.cpu(), done.cpu(),
+                                  None)
+    with torch.no_grad():
+        loss_ref, td_ref = dist_nstep_td_error(data_cpu, 0.99, -10, 10, A, nstep)
+    assert abs(loss_hip.item() - loss_ref.item()) < 1e-3
+    assert mean_rel_err(td_hip.cpu(), td_ref) < 1e-3
+
+
+def test_scatter_connection_hip_vs_oracle(ext):
+    from ding.torch_utils import ScatterConnection
+    B, M, N, H, W = 4, 32, 16, 16, 16
+    x = torch.randn(B, M, N, device="cuda")
+    loc = torch.stack(
+        [torch.randint(0, H, (B, M), device="cuda"), torch.randint(0, W, (B, M), device="cuda")], dim=-1
+    )
+    sc = ScatterConnection('add')
+    with torch.no_grad():
+        out_hip = sc(x, (H, W), loc)
+        out_ref = sc(x.cpu(), (H, W), loc.cpu())
+    assert torch.allclose(out_hip.cpu(), out_ref, atol=1e-4)
+
+
+def test_kernel_speedup_report(ext):
+    """Timing harness (DI-hpc testbase pattern): origin vs HIP over 6 iters."""
+    import time
+    from ding.rl_utils import gae, gae_data
+    T, B = 1024, 64
+    value = torch.randn(T, B, device="cuda")
+    next_value = torch.randn(T, B, device="cuda")
+    reward = torch.randn(T, B, device="cuda")
+
+    def time_fn(fn, iters=6):
+        fn()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(iters):
+            fn()
+        torch.cuda.synchronize()
+        return (time.perf_counter() - t0) / iters
+
+    with torch.no_grad():
+        t_hip = time_fn(lambda: gae(gae_data(value, next_value, reward, None, None), 0.99, 0.95))
+    import os, importlib
+    os.environ["DI_ENGINE_DISABLE_HIP"] = "1"
+    import ding.ops.dispatch as d
+    importlib.reload(d)
+    try:
+        with torch.no_grad():
+            t_ref = time_fn(lambda: gae(gae_data(value.clone(), next_value.clone(), reward, None, None), 0.99, 0.95))
+    finally:
+        os.environ.pop("DI_ENGINE_DISABLE_HIP")
+        importlib.reload(d)
+    print(f"\nGAE T={T} B={B}: hip {t_hip*1e3:.3f} ms vs eager {t_ref*1e3:.3f} ms -> {t_ref/t_hip:.1f}x")
+    assert t_hip < t_ref, "HIP scan should beat the T-step eager loop"
