@@ -2,3 +2,4 @@ from .base_policy import Policy, CommandModePolicy, create_policy, get_policy_cl
 from .common_utils import default_preprocess_learn, single_env_forward_wrapper, single_env_forward_wrapper_ttorch
 from .dqn import DQNPolicy, DQNSTDIMPolicy
 from .ppo import PPOPolicy, PPOPGPolicy, PPOOffPolicy
+from . import command_mode_policy_instance  # registers '<name>_command' variants

@@ -1,0 +1,5 @@
+from .learner.base_learner import BaseLearner, create_learner, LoadCkptHook, SaveCkptHook, LogShowHook, Hook, build_learner_hook_by_cfg
+from .collector.sample_serial_collector import SampleSerialCollector, EpisodeSerialCollector, create_serial_collector, get_serial_collector_cls, ISerialCollector
+from .collector.interaction_serial_evaluator import InteractionSerialEvaluator, create_serial_evaluator
+from .coordinator.base_serial_commander import BaseSerialCommander
+from .replay_buffer.naive_buffer import NaiveReplayBuffer, AdvancedReplayBuffer, EpisodeReplayBuffer, create_buffer, get_buffer_cls, IBuffer
