@@ -1,3 +1,4 @@
 from .q_learning import DQN, BDQ, C51DQN, QRDQN, IQN, FQF, RainbowDQN, DRQN, GTrXLDQN
 from .vac import VAC
 from .qac import ContinuousQAC, DiscreteQAC
+from .pg import PG

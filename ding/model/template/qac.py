@@ -179,6 +179,7 @@ class DiscreteQAC(nn.Module):
         activation=nn.ReLU(),
         norm_type: Optional[str] = None,
         encoder_hidden_size_list: Optional[Sequence] = None,
+        action_space: str = 'discrete',  # accepted for cfg parity; always categorical
     ):
         super().__init__()
         obs_shape, action_shape = squeeze(obs_shape), squeeze(action_shape)

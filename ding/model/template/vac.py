@@ -27,9 +27,9 @@ class VAC(nn.Module):
         action_space: str = 'discrete',
         share_encoder: bool = True,
         encoder_hidden_size_list: Sequence = [128, 128, 64],
-        actor_head_hidden_size: int = 64,
+        actor_head_hidden_size: Optional[int] = None,
         actor_head_layer_num: int = 1,
-        critic_head_hidden_size: int = 64,
+        critic_head_hidden_size: Optional[int] = None,
         critic_head_layer_num: int = 1,
         activation=nn.ReLU(),
         norm_type: Optional[str] = None,
@@ -43,6 +43,10 @@ class VAC(nn.Module):
         super().__init__()
         obs_shape, action_shape = squeeze(obs_shape), squeeze(action_shape)
         self.obs_shape, self.action_shape = obs_shape, action_shape
+        if actor_head_hidden_size is None:
+            actor_head_hidden_size = encoder_hidden_size_list[-1]
+        if critic_head_hidden_size is None:
+            critic_head_hidden_size = encoder_hidden_size_list[-1]
         self.impala_cnn_encoder = impala_cnn_encoder
         self.share_encoder = share_encoder
 

@@ -2,4 +2,9 @@ from .base_policy import Policy, CommandModePolicy, create_policy, get_policy_cl
 from .common_utils import default_preprocess_learn, single_env_forward_wrapper, single_env_forward_wrapper_ttorch
 from .dqn import DQNPolicy, DQNSTDIMPolicy
 from .ppo import PPOPolicy, PPOPGPolicy, PPOOffPolicy
+from .c51 import C51Policy, QRDQNPolicy, IQNPolicy, FQFPolicy, RainbowDQNPolicy, SQLPolicy, MDQNPolicy, BDQPolicy, SQNPolicy
+from .ddpg import DDPGPolicy, TD3Policy
+from .sac import SACPolicy, DiscreteSACPolicy, SQILSACPolicy
+from .a2c import A2CPolicy
+from .impala import IMPALAPolicy, PGPolicy
 from . import command_mode_policy_instance  # registers '<name>_command' variants

@@ -271,6 +271,11 @@ class Policy(ABC):
     def _get_on_policy(self):
         return self._on_policy
 
+    def _get_priority(self):
+        if hasattr(self, '_priority'):
+            return self._priority
+        return self._cfg.get('priority', False)
+
 
 class CommandModePolicy(Policy):
     """Policy + command mode: per-iteration hyperparameter schedule info used

@@ -36,8 +36,10 @@ def default_preprocess_learn(
         assert use_priority, "priority_IS_weight requires priority"
         if 'priority_IS' in data:
             data['weight'] = data['priority_IS']
-        else:
+        elif 'IS' in data:
             data['weight'] = data['IS']
+        else:  # buffer without IS tracking (e.g. naive): uniform weights
+            data['weight'] = None
     else:
         data['weight'] = data.get('weight', None)
 

@@ -34,6 +34,8 @@ class Adder:
         value = torch.stack([d['value'] for d in data])
         next_value = torch.stack([d['value'] for d in data][1:] + [last_value])
         reward = torch.stack([d['reward'] for d in data])
+        if reward.dim() == value.dim() + 1 and reward.shape[-1] == 1:
+            reward = reward.squeeze(-1)  # [T, 1] env reward vs scalar values
         if cuda:
             value, next_value, reward = value.cuda(), next_value.cuda(), reward.cuda()
         adv = gae(gae_data(value, next_value, reward, None, None), gamma, gae_lambda)

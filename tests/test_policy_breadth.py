@@ -1,0 +1,129 @@
+"""Smoke-level training for every registered algorithm: 1-2 serial_pipeline
+iterations on CartPole (discrete) / Pendulum (continuous), mirroring the
+reference's entry/tests/test_serial_entry.py strategy."""
+import copy
+
+import pytest
+
+from ding.entry import serial_pipeline, serial_pipeline_onpolicy
+from ding.utils import EasyDict
+
+
+def cartpole_cfg(policy_type: str, extra_policy: dict = None, buffer_type: str = 'naive') -> tuple:
+    main = EasyDict(dict(
+        exp_name=f'exp/test_{policy_type}',
+        env=dict(collector_env_num=2, evaluator_env_num=2, n_evaluator_episode=2, stop_value=195),
+        policy=dict(
+            cuda=False,
+            model=dict(obs_shape=4, action_shape=2, encoder_hidden_size_list=[32, 32]),
+            nstep=3,
+            discount_factor=0.97,
+            learn=dict(update_per_collect=2, batch_size=16, learning_rate=1e-3),
+            collect=dict(n_sample=16),
+            eval=dict(evaluator=dict(eval_freq=int(1e6))),
+            other=dict(
+                eps=dict(type='exp', start=0.95, end=0.1, decay=10000),
+                replay_buffer=dict(type=buffer_type, replay_buffer_size=1000),
+            ),
+        ),
+    ))
+    if extra_policy:
+        from ding.utils import deep_merge_dicts
+        main.policy = EasyDict(deep_merge_dicts(main.policy, extra_policy))
+    create = EasyDict(dict(
+        env=dict(type='cartpole', import_names=['dizoo.classic_control.cartpole.envs.cartpole_env']),
+        env_manager=dict(type='base'),
+        policy=dict(type=policy_type),
+    ))
+    return main, create
+
+
+def pendulum_cfg(policy_type: str, extra_policy: dict = None) -> tuple:
+    main = EasyDict(dict(
+        exp_name=f'exp/test_{policy_type}',
+        env=dict(collector_env_num=2, evaluator_env_num=2, n_evaluator_episode=2, stop_value=-200, act_scale=True),
+        policy=dict(
+            cuda=False,
+            random_collect_size=24,
+            model=dict(obs_shape=3, action_shape=1),
+            learn=dict(update_per_collect=2, batch_size=16),
+            collect=dict(n_sample=16),
+            eval=dict(evaluator=dict(eval_freq=int(1e6))),
+            other=dict(replay_buffer=dict(replay_buffer_size=1000)),
+        ),
+    ))
+    if extra_policy:
+        from ding.utils import deep_merge_dicts
+        main.policy = EasyDict(deep_merge_dicts(main.policy, extra_policy))
+    create = EasyDict(dict(
+        env=dict(type='pendulum', import_names=['dizoo.classic_control.pendulum.envs.pendulum_env']),
+        env_manager=dict(type='base'),
+        policy=dict(type=policy_type),
+    ))
+    return main, create
+
+
+@pytest.mark.parametrize('ptype', ['dqn', 'c51', 'qrdqn', 'iqn', 'rainbow', 'sql', 'mdqn', 'sqn'])
+def test_value_based_smoke(ptype):
+    extra = dict(nstep=1) if ptype in ('mdqn', 'sqn') else None
+    main, create = cartpole_cfg(ptype, extra_policy=extra)
+    serial_pipeline((main, create), seed=0, max_train_iter=2)
+
+
+def test_fqf_smoke():
+    main, create = cartpole_cfg('fqf')
+    serial_pipeline((main, create), seed=0, max_train_iter=2)
+
+
+def test_per_buffer_smoke():
+    main, create = cartpole_cfg('dqn', extra_policy=dict(priority=True, priority_IS_weight=True),
+                                buffer_type='advanced')
+    serial_pipeline((main, create), seed=0, max_train_iter=2)
+
+
+def test_a2c_smoke():
+    main, create = cartpole_cfg('a2c')
+    main.policy.learn.pop('update_per_collect', None)
+    serial_pipeline_onpolicy((main, create), seed=0, max_train_iter=2)
+
+
+def test_impala_smoke():
+    main, create = cartpole_cfg('impala', extra_policy=dict(unroll_len=8, learn=dict(batch_size=2)))
+    serial_pipeline((main, create), seed=0, max_train_iter=2)
+
+
+def test_ppo_offpolicy_smoke():
+    main, create = cartpole_cfg(
+        'ppo_offpolicy',
+        extra_policy=dict(model=dict(action_space='discrete'), learn=dict(epoch_per_collect=1))
+    )
+    serial_pipeline((main, create), seed=0, max_train_iter=2)
+
+
+def test_ddpg_smoke():
+    main, create = pendulum_cfg(
+        'ddpg', extra_policy=dict(model=dict(action_space='regression'),
+                                  learn=dict(learning_rate_actor=1e-3, learning_rate_critic=1e-3))
+    )
+    serial_pipeline((main, create), seed=0, max_train_iter=2)
+
+
+def test_td3_smoke():
+    main, create = pendulum_cfg(
+        'td3', extra_policy=dict(model=dict(action_space='regression', twin_critic=True),
+                                 learn=dict(learning_rate_actor=1e-3, learning_rate_critic=1e-3))
+    )
+    serial_pipeline((main, create), seed=0, max_train_iter=2)
+
+
+def test_sac_smoke():
+    main, create = pendulum_cfg(
+        'sac', extra_policy=dict(model=dict(action_space='reparameterization', twin_critic=True))
+    )
+    serial_pipeline((main, create), seed=0, max_train_iter=2)
+
+
+def test_discrete_sac_smoke():
+    main, create = cartpole_cfg('discrete_sac', extra_policy=dict(model=dict(twin_critic=True)))
+    main.policy.random_collect_size = 0
+    serial_pipeline((main, create), seed=0, max_train_iter=2)
