@@ -2,3 +2,6 @@ from .q_learning import DQN, BDQ, C51DQN, QRDQN, IQN, FQF, RainbowDQN, DRQN, GTr
 from .vac import VAC
 from .qac import ContinuousQAC, DiscreteQAC
 from .pg import PG
+from .qmix import QMix, Mixer
+from .coma_model import COMA
+from .mavac import MAVAC

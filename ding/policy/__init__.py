@@ -7,4 +7,5 @@ from .ddpg import DDPGPolicy, TD3Policy
 from .sac import SACPolicy, DiscreteSACPolicy, SQILSACPolicy
 from .a2c import A2CPolicy
 from .impala import IMPALAPolicy, PGPolicy
+from .qmix import QMIXPolicy, WQMIXPolicy, COMAPolicy
 from . import command_mode_policy_instance  # registers '<name>_command' variants

@@ -105,8 +105,9 @@ class PPOPolicy(Policy):
         data = default_preprocess_learn(data, ignore_done=self._cfg.learn.ignore_done, use_nstep=False)
         if self._cuda:
             data = to_device(data, self._device)
-        data['obs'] = data['obs'].float()
-        if 'next_obs' in data:
+        if isinstance(data['obs'], torch.Tensor):
+            data['obs'] = data['obs'].float()
+        if 'next_obs' in data and isinstance(data['next_obs'], torch.Tensor):
             data['next_obs'] = data['next_obs'].float()
         self._learn_model.train()
         return_infos = []

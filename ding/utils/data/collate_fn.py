@@ -70,17 +70,24 @@ def timestep_collate(batch: List[Dict[str, Any]]) -> Dict[str, Any]:
     if 'prev_state' in elem:
         # [B][T] -> [T][B]
         prev_state = list(zip(*[d.pop('prev_state') for d in batch]))
-    out = {}
-    for key in elem:
-        vals = [d[key] for d in batch]
+    def _stack_time(vals):
+        """vals: B samples, each a T-list (or [T,...] tensor) -> [T, B, ...]."""
+        if isinstance(vals[0], dict):
+            return {k: _stack_time([v[k] for v in vals]) for k in vals[0]}
         if isinstance(vals[0], (list, tuple)):
             stacked = [default_collate(list(ts_vals)) for ts_vals in zip(*vals)]  # per timestep over batch
-            out[key] = torch.stack([s if isinstance(s, torch.Tensor) else torch.as_tensor(s) for s in stacked], 0)
-        elif isinstance(vals[0], torch.Tensor) and vals[0].dim() >= 1:
-            # already [T, ...] per sample
-            out[key] = torch.stack(vals, 1)
-        else:
-            out[key] = default_collate(vals)
+            if isinstance(stacked[0], torch.Tensor):
+                return torch.stack(stacked, 0)
+            if isinstance(stacked[0], dict):
+                return {k: torch.stack([s[k] for s in stacked], 0) for k in stacked[0]}
+            return stacked
+        if isinstance(vals[0], torch.Tensor) and vals[0].dim() >= 1:
+            return torch.stack(vals, 1)  # already [T, ...] per sample
+        return default_collate(vals)
+
+    out = {}
+    for key in elem:
+        out[key] = _stack_time([d[key] for d in batch])
     if prev_state is not None:
         out['prev_state'] = [list(s) for s in prev_state]
     return out

@@ -162,15 +162,16 @@ def split_data_generator(data: dict, split_size: int, shuffle: bool = True):
         indices = torch.randperm(length, device="cpu")
     else:
         indices = torch.arange(length)
+    def _index(v, idx):
+        if isinstance(v, torch.Tensor) and v.dim() > 0 and v.shape[0] == length:
+            return v[idx.to(v.device)]
+        if isinstance(v, dict):
+            return {k: _index(sub, idx) for k, sub in v.items()}
+        return v
+
     for start in range(0, length - split_size + 1, split_size):
         idx = indices[start:start + split_size]
-        batch = {}
-        for k, v in data.items():
-            if isinstance(v, torch.Tensor) and v.dim() > 0 and v.shape[0] == length:
-                batch[k] = v[idx.to(v.device)]
-            else:
-                batch[k] = v
-        yield batch
+        yield {k: _index(v, idx) for k, v in data.items()}
 
 
 def flatten_dict(data: dict, delimiter: str = "/", prefix: str = "") -> dict:

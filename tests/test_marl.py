@@ -1,0 +1,56 @@
+"""MARL smoke tests (QMIX/WQMIX/COMA/MAPPO on the cooperative matrix env)."""
+import pytest
+from ding.entry import serial_pipeline, serial_pipeline_onpolicy
+from ding.utils import EasyDict
+
+
+def _marl_cfg(ptype, model, extra=None):
+    main = EasyDict(dict(
+        exp_name=f'exp/test_{ptype}',
+        env=dict(collector_env_num=2, evaluator_env_num=2, n_evaluator_episode=2, stop_value=100),
+        policy=dict(
+            cuda=False,
+            model=model,
+            learn=dict(update_per_collect=1, batch_size=4, learning_rate=5e-4),
+            collect=dict(n_sample=8, unroll_len=5, env_num=2),
+            eval=dict(env_num=2, evaluator=dict(eval_freq=int(1e6))),
+            other=dict(eps=dict(type='exp', start=1, end=0.05, decay=10000),
+                       replay_buffer=dict(replay_buffer_size=200)),
+        ),
+    ))
+    if extra:
+        from ding.utils import deep_merge_dicts
+        main.policy = EasyDict(deep_merge_dicts(main.policy, extra))
+    create = EasyDict(dict(
+        env=dict(type='coop_matrix', import_names=['dizoo.multiagent.envs.coop_matrix_env']),
+        env_manager=dict(type='base'),
+        policy=dict(type=ptype),
+    ))
+    return main, create
+
+
+QMIX_MODEL = dict(agent_num=3, obs_shape=8, global_obs_shape=12, action_shape=4, hidden_size_list=[32, 32],
+                  mixer=True)
+
+
+def test_qmix():
+    serial_pipeline(_marl_cfg('qmix', QMIX_MODEL), seed=0, max_train_iter=2)
+
+
+def test_wqmix():
+    serial_pipeline(_marl_cfg('wqmix', QMIX_MODEL), seed=0, max_train_iter=2)
+
+
+def test_coma():
+    model = dict(agent_num=3, obs_shape=dict(agent_state=8, global_state=12), action_shape=4)
+    serial_pipeline(_marl_cfg('coma', model), seed=0, max_train_iter=2)
+
+
+def test_mappo():
+    model = dict(agent_obs_shape=8, global_obs_shape=12, action_shape=4, agent_num=3,
+                 actor_hidden_size_list=[32, 32], critic_hidden_size_list=[32, 32])
+    extra = dict(multi_agent=True, action_space='discrete',
+                 learn=dict(epoch_per_collect=2, batch_size=8, learning_rate=3e-4),
+                 collect=dict(n_sample=32, unroll_len=1, discount_factor=0.99, gae_lambda=0.95))
+    main, create = _marl_cfg('ppo', model, extra)
+    serial_pipeline_onpolicy((main, create), seed=0, max_train_iter=2)
