@@ -5,3 +5,4 @@ from .pg import PG
 from .qmix import QMix, Mixer
 from .coma_model import COMA
 from .mavac import MAVAC
+from .bc import DiscreteBC, ContinuousBC, EDAC

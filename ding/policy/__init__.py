@@ -8,4 +8,5 @@ from .sac import SACPolicy, DiscreteSACPolicy, SQILSACPolicy
 from .a2c import A2CPolicy
 from .impala import IMPALAPolicy, PGPolicy
 from .qmix import QMIXPolicy, WQMIXPolicy, COMAPolicy
+from .offline import BehaviourCloningPolicy, CQLPolicy, DiscreteCQLPolicy, TD3BCPolicy, IQLPolicy, EDACPolicy
 from . import command_mode_policy_instance  # registers '<name>_command' variants
