@@ -6,3 +6,4 @@ from .qmix import QMix, Mixer
 from .coma_model import COMA
 from .mavac import MAVAC
 from .bc import DiscreteBC, ContinuousBC, EDAC
+from .decision_transformer import DecisionTransformer

@@ -9,4 +9,6 @@ from .a2c import A2CPolicy
 from .impala import IMPALAPolicy, PGPolicy
 from .qmix import QMIXPolicy, WQMIXPolicy, COMAPolicy
 from .offline import BehaviourCloningPolicy, CQLPolicy, DiscreteCQLPolicy, TD3BCPolicy, IQLPolicy, EDACPolicy
+from .r2d2 import R2D2Policy
+from .dt import DTPolicy
 from . import command_mode_policy_instance  # registers '<name>_command' variants

@@ -31,10 +31,11 @@ def lists_to_dicts(data: Sequence[Union[dict, tuple]], recursive: bool = False) 
     first = data[0]
     if isinstance(first, dict):
         keys = first.keys()
+        _no_recurse = ('prev_state', 'prev_actor_state', 'prev_critic_state')
         out = {}
         for k in keys:
             vals = [d[k] for d in data]
-            if recursive and isinstance(vals[0], dict):
+            if recursive and isinstance(vals[0], dict) and k not in _no_recurse:
                 vals = lists_to_dicts(vals, recursive=True)
             out[k] = vals
         if isinstance(first, EasyDict):

@@ -127,3 +127,17 @@ def test_discrete_sac_smoke():
     main, create = cartpole_cfg('discrete_sac', extra_policy=dict(model=dict(twin_critic=True)))
     main.policy.random_collect_size = 0
     serial_pipeline((main, create), seed=0, max_train_iter=2)
+
+
+def test_r2d2_smoke():
+    main, create = cartpole_cfg(
+        'r2d2',
+        extra_policy=dict(
+            priority=True, priority_IS_weight=True, nstep=2, burnin_step=2, learn_unroll_len=6,
+            model=dict(obs_shape=4, action_shape=2, encoder_hidden_size_list=[32, 32]),
+            learn=dict(update_per_collect=1, batch_size=4, learning_rate=1e-4),
+            collect=dict(n_sample=8, env_num=2), eval=dict(env_num=2, evaluator=dict(eval_freq=int(1e6))),
+        ),
+        buffer_type='advanced',
+    )
+    serial_pipeline((main, create), seed=0, max_train_iter=2)
