@@ -68,12 +68,17 @@ class TCPMQ(MQ):
         raise ConnectionError(f"cannot dial peer {addr}")
 
     def _accept_loop(self):
+        n_accepted = 0
         while not self._end:
             try:
-                conn, _ = self._server.accept()
+                conn, addr = self._server.accept()
             except OSError:
                 break
             conn.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+            # bus semantics: accepted links are full-duplex publish targets too
+            with self._peer_lock:
+                self._peers[f"accepted://{addr[0]}:{addr[1]}:{n_accepted}"] = conn
+            n_accepted += 1
             t = threading.Thread(target=self._recv_loop, args=(conn, ), daemon=True)
             t.start()
             self._threads.append(t)

@@ -37,7 +37,7 @@ def ddp_termination_checker(max_env_step: Optional[int] = None, max_train_iter: 
         max_train_iter = np.inf
 
     def _check(ctx):
-        import ding.framework.task as _t
+        from ding.framework import task as _task
         if rank == 0:
             if getattr(ctx, 'env_step', 0) >= max_env_step or getattr(ctx, 'train_iter', 0) >= max_train_iter:
                 finish = torch.ones(1, dtype=torch.int64)

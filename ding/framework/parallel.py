@@ -131,6 +131,7 @@ class Parallel:
         self.node_id = node_id
         self.attach_to = attach_to
         self.labels = labels or set()
+        import ding.framework.message_queue  # populate MQ_REGISTRY in spawned procs
         self._mq = MQ_REGISTRY.get(mq_type)(listen_to=listen_to, attach_to=attach_to)
         self._mq.listen()
         self.is_active = True
