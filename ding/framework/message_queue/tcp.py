@@ -62,6 +62,10 @@ class TCPMQ(MQ):
                 s.connect((host, port))
                 with self._peer_lock:
                     self._peers[addr] = s
+                # full duplex: peers publish back over the same link
+                t = threading.Thread(target=self._recv_loop, args=(s, ), daemon=True)
+                t.start()
+                self._threads.append(t)
                 return
             except (ConnectionRefusedError, OSError):
                 time.sleep(wait)
