@@ -1,0 +1,6 @@
+from .base_world_model import (
+    WorldModel, DynaWorldModel, DreamWorldModel, HybridWorldModel, create_world_model,
+    get_rollout_length_scheduler,
+)
+from .mbpo import MBPOWorldModel, EnsembleModel, EnsembleFC
+from .ddppo import DDPPOWorldMode
