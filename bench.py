@@ -26,6 +26,9 @@ import time
 
 import torch
 
+# MIOpen autotune: pick the fastest conv algorithms for the fixed bench shapes
+torch.backends.cudnn.benchmark = True
+
 
 def build_ppo_policy(device: str, multi_gpu: bool):
     from ding.policy import PPOPolicy

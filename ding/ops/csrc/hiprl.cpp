@@ -10,6 +10,16 @@ torch::Tensor c51_project(
     double v_max, double gamma_n
 );
 torch::Tensor scatter_connection(torch::Tensor x, torch::Tensor index, int64_t H, int64_t W, int64_t scatter_add);
+std::vector<torch::Tensor> ppo_fwd(
+    torch::Tensor logit_new, torch::Tensor logit_old, torch::Tensor action, torch::Tensor value_new,
+    torch::Tensor value_old, torch::Tensor adv, torch::Tensor ret, torch::Tensor weight, double clip_ratio,
+    int64_t use_value_clip
+);
+std::vector<torch::Tensor> ppo_bwd(
+    torch::Tensor logit_new, torch::Tensor action, torch::Tensor value_new, torch::Tensor value_old,
+    torch::Tensor adv, torch::Tensor ret, torch::Tensor weight, torch::Tensor fwd_out, double clip_ratio,
+    double grad_policy, double grad_value, double grad_entropy
+);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "DI-engine MI355X HIP kernels (gfx950)";
@@ -17,4 +27,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("multistep_forward_view", &multistep_forward_view, "TD(lambda) forward-view reverse scan");
     m.def("c51_project", &c51_project, "C51 categorical projection");
     m.def("scatter_connection", &scatter_connection, "entity->spatial scatter");
+    m.def("ppo_fwd", &ppo_fwd, "fused PPO loss forward");
+    m.def("ppo_bwd", &ppo_bwd, "fused PPO loss backward");
 }

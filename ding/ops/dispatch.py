@@ -39,6 +39,19 @@ def hip_ops():
 _DISABLED = os.environ.get("DI_ENGINE_DISABLE_HIP", "0") in ("1", "true", "True")
 
 
+def use_hip_autograd(x: torch.Tensor) -> bool:
+    """Lane check for fused ops that implement their own backward (gradient
+    through ``x`` is fine)."""
+    if _DISABLED or not isinstance(x, torch.Tensor) or not x.is_cuda:
+        return False
+    if not is_available():
+        raise RuntimeError(
+            f"ding.ops HIP extension not built but got a GPU tensor (import error: {_EXT_ERR}). "
+            "Run `python setup_ops.py` or set DI_ENGINE_DISABLE_HIP=1."
+        )
+    return True
+
+
 def use_hip(x: torch.Tensor) -> bool:
     """Decide the lane for this call. Raises on GPU-without-extension.
 
@@ -83,6 +96,49 @@ def multistep_forward_view(bootstrap_values, rewards, gammas, lambda_, done):
         bootstrap_values.contiguous().float(), rewards.contiguous().float(), gammas.contiguous().float(),
         lambda_.contiguous().float(), done.contiguous().float()
     ).to(rewards.dtype)
+
+
+class _FusedPPODiscrete(torch.autograd.Function):
+    """Fused discrete PPO loss: forward packs the per-row terms in one kernel,
+    backward produces analytic d(logit_new)/d(value_new) in one kernel."""
+
+    @staticmethod
+    def forward(ctx, logit_new, value_new, logit_old, action, value_old, adv, ret, weight, clip_ratio,
+                use_value_clip):
+        ext = _load()
+        w = weight if weight is not None else torch.empty(0, device=logit_new.device)
+        out = ext.ppo_fwd(
+            logit_new.contiguous(), logit_old.contiguous(), action.contiguous(), value_new.contiguous(),
+            value_old.contiguous(), adv.contiguous(), ret.contiguous(), w, float(clip_ratio), int(use_value_clip)
+        )[0]
+        ctx.save_for_backward(logit_new, action, value_new, value_old, adv, ret, w, out)
+        ctx.clip_ratio = float(clip_ratio)
+        policy_loss = out[:, 0].mean()
+        value_loss = out[:, 1].mean()
+        entropy_loss = out[:, 2].mean()
+        approx_kl = out[:, 3].mean()
+        clipfrac = out[:, 4].mean()
+        ctx.mark_non_differentiable(approx_kl, clipfrac)
+        return policy_loss, value_loss, entropy_loss, approx_kl, clipfrac
+
+    @staticmethod
+    def backward(ctx, g_policy, g_value, g_entropy, g_kl=None, g_clip=None):
+        ext = _load()
+        logit_new, action, value_new, value_old, adv, ret, w, out = ctx.saved_tensors
+        d_logit, d_value = ext.ppo_bwd(
+            logit_new.contiguous(), action.contiguous(), value_new.contiguous(), value_old.contiguous(),
+            adv.contiguous(), ret.contiguous(), w, out, ctx.clip_ratio,
+            float(g_policy), float(g_value), float(g_entropy)
+        )
+        return d_logit, d_value, None, None, None, None, None, None, None, None
+
+
+def fused_ppo_error(logit_new, logit_old, action, value_new, value_old, adv, ret, weight,
+                    clip_ratio: float, use_value_clip: bool):
+    """Returns (policy_loss, value_loss, entropy_loss, approx_kl, clipfrac)."""
+    return _FusedPPODiscrete.apply(
+        logit_new, value_new, logit_old, action, value_old, adv, ret, weight, clip_ratio, use_value_clip
+    )
 
 
 def scatter_connection(x, index, H: int, W: int, scatter_type: str):

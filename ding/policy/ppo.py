@@ -114,8 +114,15 @@ class PPOPolicy(Policy):
         for epoch in range(self._cfg.learn.epoch_per_collect):
             if self._recompute_adv:
                 with torch.no_grad():
-                    value = self._learn_model.forward(data['obs'], mode='compute_critic')['value']
-                    next_value = self._learn_model.forward(data['next_obs'], mode='compute_critic')['value']
+                    if isinstance(data['obs'], torch.Tensor):
+                        # one batched critic pass over [obs; next_obs] — halves
+                        # the number of large forwards vs two separate calls
+                        both = torch.cat([data['obs'], data['next_obs']], dim=0)
+                        values = self._learn_model.forward(both, mode='compute_critic')['value']
+                        value, next_value = values.chunk(2, dim=0)
+                    else:
+                        value = self._learn_model.forward(data['obs'], mode='compute_critic')['value']
+                        next_value = self._learn_model.forward(data['next_obs'], mode='compute_critic')['value']
                     if self._value_norm:
                         value *= float(self._running_mean_std.std[0])
                         next_value *= float(self._running_mean_std.std[0])

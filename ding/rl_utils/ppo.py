@@ -110,8 +110,21 @@ def ppo_error(
     kl_type: str = 'k1',
 ) -> Tuple[namedtuple, namedtuple]:
     """Combined PPO loss for discrete actions. See ppo_data for fields."""
+    from ding.ops import dispatch
     logit_new, logit_old, action, value_new, value_old, adv, return_, weight = data[:8]
     logit_pretrained = data[8] if len(data) > 8 else None
+    if (
+        dual_clip is None and logit_pretrained is None and isinstance(logit_new, torch.Tensor)
+        and logit_new.dim() == 2 and action.dim() == 1 and logit_new.dtype == torch.float32
+        and dispatch.use_hip_autograd(logit_new)
+    ):
+        policy_loss, value_loss, entropy_loss, approx_kl, clipfrac = dispatch.fused_ppo_error(
+            logit_new, logit_old.detach(), action, value_new, value_old.detach(), adv.detach(), return_.detach(),
+            None if weight is None else weight.detach().float(), clip_ratio, use_value_clip
+        )
+        kl_div = torch.zeros((), dtype=policy_loss.dtype, device=policy_loss.device)
+        return ppo_loss(policy_loss, value_loss, entropy_loss, kl_div), \
+            ppo_info(float(approx_kl), float(clipfrac))
     pol, info = ppo_policy_error(
         ppo_policy_data(logit_new, logit_old, action, adv, weight, logit_pretrained), clip_ratio, dual_clip,
         kl_type=kl_type

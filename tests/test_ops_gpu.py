@@ -158,3 +158,48 @@ def test_kernel_speedup_report(ext):
         importlib.reload(d)
     print(f"\nGAE T={T} B={B}: hip {t_hip*1e3:.3f} ms vs eager {t_ref*1e3:.3f} ms -> {t_ref/t_hip:.1f}x")
     assert t_hip < t_ref, "HIP scan should beat the T-step eager loop"
+
+
+def test_fused_ppo_vs_oracle(ext):
+    """Fused PPO loss forward+backward vs the eager fp32 oracle."""
+    import os, importlib
+    from ding.rl_utils import ppo_data, ppo_error
+    torch.manual_seed(0)
+    B, N = 320, 6
+    logit_new = torch.randn(B, N, device="cuda", requires_grad=True)
+    value_new = torch.randn(B, device="cuda", requires_grad=True)
+    logit_old = torch.randn(B, N, device="cuda")
+    action = torch.randint(0, N, (B, ), device="cuda")
+    value_old = torch.randn(B, device="cuda")
+    adv = torch.randn(B, device="cuda")
+    ret = torch.randn(B, device="cuda")
+    data = ppo_data(logit_new, logit_old, action, value_new, value_old, adv, ret, None)
+    loss, info = ppo_error(data)
+    total = loss.policy_loss + 0.5 * loss.value_loss - 0.01 * loss.entropy_loss
+    total.backward()
+    g_logit_hip = logit_new.grad.clone()
+    g_value_hip = value_new.grad.clone()
+
+    # oracle on the eager lane
+    os.environ["DI_ENGINE_DISABLE_HIP"] = "1"
+    import ding.ops.dispatch as d
+    importlib.reload(d)
+    try:
+        logit_new2 = logit_new.detach().clone().requires_grad_(True)
+        value_new2 = value_new.detach().clone().requires_grad_(True)
+        data2 = ppo_data(logit_new2, logit_old, action, value_new2, value_old, adv, ret, None)
+        loss2, info2 = ppo_error(data2)
+        total2 = loss2.policy_loss + 0.5 * loss2.value_loss - 0.01 * loss2.entropy_loss
+        total2.backward()
+    finally:
+        os.environ.pop("DI_ENGINE_DISABLE_HIP")
+        importlib.reload(d)
+    assert abs(loss.policy_loss.item() - loss2.policy_loss.item()) < 1e-4
+    assert abs(loss.value_loss.item() - loss2.value_loss.item()) < 1e-4
+    assert abs(loss.entropy_loss.item() - loss2.entropy_loss.item()) < 1e-4
+    assert abs(info.approx_kl - info2.approx_kl) < 1e-4
+    assert abs(info.clipfrac - info2.clipfrac) < 1e-4
+    assert torch.allclose(g_logit_hip, logit_new2.grad, atol=1e-5), \
+        f"max logit grad err {(g_logit_hip - logit_new2.grad).abs().max()}"
+    assert torch.allclose(g_value_hip, value_new2.grad, atol=1e-5), \
+        f"max value grad err {(g_value_hip - value_new2.grad).abs().max()}"
