@@ -19,6 +19,16 @@ from ding.utils import REWARD_MODEL_REGISTRY, EasyDict
 from .base_reward_model import BaseRewardModel
 
 
+
+
+def _rewrite_reward(item: dict, value: float) -> None:
+    """Overwrite the reward while preserving its original shape (n-step
+    reward vectors stay n-step vectors)."""
+    old = torch.as_tensor(item['reward'], dtype=torch.float32)
+    if old.dim() == 0:
+        old = old.reshape(1)
+    item['reward'] = torch.full_like(old, float(value))
+
 def _load_expert(path: str) -> list:
     with open(path, 'rb') as f:
         return pickle.load(f)
@@ -93,7 +103,7 @@ class GailRewardModel(BaseRewardModel):
             d_prob = torch.sigmoid(self.reward_model(x)).squeeze(-1)
             reward = -torch.log(1 - d_prob + 1e-8)
         for item, r in zip(out, reward):
-            item['reward'] = torch.as_tensor([r.item()])
+            _rewrite_reward(item, r.item())
         return out
 
 
@@ -161,7 +171,7 @@ class GuidedCostRewardModel(BaseRewardModel):
         with torch.no_grad():
             reward = -self.reward_model(x).squeeze(-1)
         for item, r in zip(out, reward):
-            item['reward'] = torch.as_tensor([r.item()])
+            _rewrite_reward(item, r.item())
         return out
 
 
@@ -208,7 +218,7 @@ class PwilRewardModel(BaseRewardModel):
             ])
             dist = (exp - x).norm(dim=1).min()
             reward = self.cfg.alpha * torch.exp(-self.cfg.beta * dist)
-            item['reward'] = torch.as_tensor([reward.item()])
+            _rewrite_reward(item, reward.item())
         return out
 
 
@@ -266,7 +276,7 @@ class RedRewardModel(BaseRewardModel):
             err = (self.predictor(x) - self.target(x)).pow(2).mean(1)
             reward = torch.exp(-self.cfg.sigma * err)
         for item, r in zip(out, reward):
-            item['reward'] = torch.as_tensor([r.item()])
+            _rewrite_reward(item, r.item())
         return out
 
 
@@ -309,7 +319,7 @@ class PdeilRewardModel(BaseRewardModel):
             x = torch.as_tensor(item['obs'], dtype=torch.float32).reshape(-1)
             d = x - self._mean
             maha = (d @ self._cov_inv @ d).clamp(min=0)
-            item['reward'] = torch.as_tensor([torch.exp(-0.5 * maha).item()])
+            _rewrite_reward(item, torch.exp(-0.5 * maha).item())
         return out
 
 
@@ -390,7 +400,7 @@ class TrexRewardModel(BaseRewardModel):
         with torch.no_grad():
             reward = self.reward_model.net(obs).squeeze(-1)
         for item, r in zip(out, reward):
-            item['reward'] = torch.as_tensor([r.item()])
+            _rewrite_reward(item, r.item())
         return out
 
 

@@ -106,7 +106,15 @@ class DynaWorldModel(WorldModel, ABC):
         img_batch = batch_size - env_batch
         env_data = env_buffer.sample(env_batch, train_iter) or []
         img_data = img_buffer.sample(img_batch, train_iter) or []
-        return list(env_data) + list(img_data)
+        data = list(env_data) + list(img_data)
+        if not data:
+            return data
+        # imagined transitions carry fewer meta keys than real ones; project
+        # onto the common key set so collation stays uniform
+        common = set(data[0].keys())
+        for d in data[1:]:
+            common &= set(d.keys())
+        return [{k: d[k] for k in common} for d in data]
 
     def fill_img_buffer(self, policy, env_buffer, img_buffer, envstep: int, train_iter: int) -> None:
         rollout_length = self.rollout_length_scheduler(envstep)

@@ -1,0 +1,75 @@
+"""Entry-layer tests: demo collection round-trip, eval entry, reward-model /
+SQIL / GAIL / DQFD pipelines, MBRL dyna pipeline, CLI import."""
+import copy
+import pickle
+
+import pytest
+import torch
+
+from ding.utils import EasyDict, deep_merge_dicts
+from tests.test_policy_breadth import cartpole_cfg, pendulum_cfg
+
+
+def test_collect_demo_and_eval(tmp_path):
+    from ding.entry import collect_demo_data, eval as eval_entry
+    main, create = cartpole_cfg('dqn')
+    path = str(tmp_path / 'demo.pkl')
+    data = collect_demo_data((main, create), seed=0, collect_count=32, expert_data_path=path)
+    assert len(data) == 32
+    with open(path, 'rb') as f:
+        loaded = pickle.load(f)
+    assert len(loaded) == 32
+    main2, create2 = cartpole_cfg('dqn')
+    value = eval_entry((main2, create2), seed=0)
+    assert isinstance(value, float)
+
+
+def test_reward_model_pipeline():
+    from ding.entry import serial_pipeline_reward_model
+    main, create = cartpole_cfg('dqn')
+    main.reward_model = EasyDict(dict(type='rnd', obs_shape=4, hidden_size_list=[16, 16], update_per_collect=1))
+    serial_pipeline_reward_model((main, create), seed=0, max_train_iter=2)
+
+
+def test_gail_pipeline(tmp_path):
+    from ding.entry import serial_pipeline_gail, collect_demo_data
+    main, create = cartpole_cfg('dqn')
+    expert = collect_demo_data((copy.deepcopy(main), copy.deepcopy(create)), seed=0, collect_count=16)
+    main.reward_model = EasyDict(dict(type='gail', input_size=5, hidden_size=16, update_per_collect=1))
+    serial_pipeline_gail((main, create), expert, seed=0, max_train_iter=2)
+
+
+def test_sqil_pipeline():
+    from ding.entry import serial_pipeline_sqil
+    main, create = cartpole_cfg('sql', extra_policy=dict(nstep=1))
+    em, ec = cartpole_cfg('sql', extra_policy=dict(nstep=1))
+    serial_pipeline_sqil((main, create), (em, ec), seed=0, max_train_iter=2)
+
+
+def test_dqfd_pipeline():
+    from ding.entry import serial_pipeline_dqfd, collect_demo_data
+    main, create = cartpole_cfg('dqn')
+    expert = collect_demo_data((copy.deepcopy(main), copy.deepcopy(create)), seed=0, collect_count=16)
+    main2, create2 = cartpole_cfg('dqn')
+    serial_pipeline_dqfd((main2, create2), expert, seed=0, max_train_iter=2)
+
+
+def test_dyna_pipeline():
+    from ding.entry import serial_pipeline_dyna
+    main, create = pendulum_cfg(
+        'sac', extra_policy=dict(model=dict(action_space='reparameterization', twin_critic=True))
+    )
+    main.world_model = EasyDict(dict(
+        type='mbpo', train_freq=8, eval_freq=100, cuda=False,
+        model=dict(state_size=3, action_size=1, hidden_size=16, ensemble_size=2, elite_size=1, batch_size=32),
+        other=dict(real_ratio=0.5, rollout_batch_size=16,
+                   imagination_buffer=dict(replay_buffer_size=1000)),
+        rollout_length_scheduler=dict(rollout_start_step=0, rollout_end_step=100, rollout_length_min=1,
+                                      rollout_length_max=2),
+    ))
+    serial_pipeline_dyna((main, create), seed=0, max_train_iter=2)
+
+
+def test_cli_imports():
+    from ding.entry import cli, cli_ditask
+    assert callable(cli) and callable(cli_ditask)
