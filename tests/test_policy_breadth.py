@@ -141,3 +141,26 @@ def test_r2d2_smoke():
         buffer_type='advanced',
     )
     serial_pipeline((main, create), seed=0, max_train_iter=2)
+
+
+def test_ppg_smoke():
+    main, create = cartpole_cfg('ppg_offpolicy', extra_policy=dict(
+        model=dict(action_space='discrete'),
+        learn=dict(epoch_per_collect=1, aux_freq=1, aux_train_epoch=1, batch_size=16)))
+    serial_pipeline((main, create), seed=0, max_train_iter=3)
+
+
+def test_acer_smoke():
+    main, create = cartpole_cfg('acer', extra_policy=dict(unroll_len=8, learn=dict(batch_size=2)))
+    serial_pipeline((main, create), seed=0, max_train_iter=2)
+
+
+def test_dqfd_policy_smoke():
+    main, create = cartpole_cfg('dqfd', extra_policy=dict(nstep=3))
+    serial_pipeline((main, create), seed=0, max_train_iter=2)
+
+
+def test_d4pg_smoke():
+    main, create = pendulum_cfg('d4pg', extra_policy=dict(
+        nstep=2, model=dict(action_space='regression', v_min=-100, v_max=100, n_atom=51)))
+    serial_pipeline((main, create), seed=0, max_train_iter=2)
