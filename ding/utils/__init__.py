@@ -30,3 +30,10 @@ from .import_helper import import_module, try_import_ceph, try_import_redis
 from .scheduler_helper import Scheduler
 from .normalizer_helper import DatasetNormalizer
 from .fast_copy import fast_copy
+from .autolog import LoggedModel, LoggedValue, TickMonitor, TimeMode, NaturalTime, TickTime, TimeProxy
+from .profiler_helper import Profiler
+from .memory_helper import SimpleMemoryProfiler
+from .loader import Loader, LoaderError, is_type, to_type, interval, enum, item, dict_, collection, optional, check_only
+from .misc_helpers import (
+    get_vi_sequence, render, K8sLauncher, generate_slurm_script, find_free_port, node_to_partition, node_to_host,
+)

@@ -1,2 +1,3 @@
 from .collate_fn import default_collate, timestep_collate, diff_shape_collate, default_decollate, ttorch_collate
 from .dataset import NaiveRLDataset, HDF5Dataset, D4RLDataset, D4RLTrajectoryDataset, create_dataset, offline_data_save_type
+from .dataloader import AsyncDataLoader

@@ -1,0 +1,88 @@
+"""Aux utils: autolog, loader DSL, profiler, memory profiler, dataloader,
+scheduler, normalizer, k8s/slurm generation."""
+import numpy as np
+import pytest
+import torch
+
+from ding.utils import (
+    LoggedModel, LoggedValue, TickTime, Loader, LoaderError, is_type, interval, enum, dict_, collection, optional,
+    Profiler, SimpleMemoryProfiler, Scheduler, DatasetNormalizer, K8sLauncher, generate_slurm_script, EasyDict,
+)
+
+
+def test_autolog():
+    class Mon(LoggedModel):
+        v = LoggedValue(float)
+
+        def __init__(self, t, expire):
+            super().__init__(t, expire)
+
+    t = TickTime()
+    m = Mon(t, expire=5)
+    for i in range(10):
+        m.v = float(i)
+        t.step()
+    vals = [v for _, v in m.range_values('v')]
+    assert len(vals) <= 6  # window pruned
+    assert m.avg('v') > 0 and m.max('v') == 9.0
+
+
+def test_loader_dsl():
+    pos_int = is_type(int) & interval(0, None)
+    assert pos_int(5) == 5
+    with pytest.raises(Exception):
+        pos_int(-1)
+    with pytest.raises(Exception):
+        pos_int("x")
+    color = enum('red', 'green')
+    assert color('red') == 'red'
+    schema = dict_(lr=is_type(float) & interval(0, 1), n=is_type(int))
+    out = schema({'lr': 0.1, 'n': 3, 'extra': 1})
+    assert out == {'lr': 0.1, 'n': 3}
+    assert collection(is_type(int))([1, 2]) == [1, 2]
+    assert optional(is_type(int))(None) is None
+    piped = is_type(int) | is_type(float)
+    assert piped(0.5) == 0.5
+
+
+def test_profilers(tmp_path):
+    p = Profiler()
+    p.mkdir(str(tmp_path / 'prof'))
+    net = torch.nn.Sequential(torch.nn.Linear(4, 8), torch.nn.ReLU(), torch.nn.Linear(8, 2))
+    mp = SimpleMemoryProfiler(net, log_folder=str(tmp_path / 'mem'), total_steps=1)
+    net(torch.randn(3, 4))
+    mp.step()
+    assert (tmp_path / 'mem' / 'params.txt').exists()
+    assert (tmp_path / 'mem' / 'activations.txt').exists()
+
+
+def test_async_dataloader():
+    from ding.utils.data import AsyncDataLoader
+    def source(bs):
+        return [{'x': torch.randn(3), 'y': 1} for _ in range(bs)]
+    loader = AsyncDataLoader(source, batch_size=4, num_workers=1)
+    batch = next(loader)
+    assert batch['x'].shape == (4, 3)
+    loader.close()
+
+
+def test_scheduler_and_normalizer():
+    s = Scheduler(EasyDict(dict(schedule_mode='reduce', factor=0.1, change_range=[0, 1], patience=1)))
+    param = 0.5
+    for _ in range(5):
+        param = s.step(10.0, param)  # flat metric -> reduce after patience
+    assert param < 0.5
+    data = {'obs': np.random.randn(100, 4).astype(np.float32)}
+    norm = DatasetNormalizer(data, 'gaussian')
+    z = norm.normalize(data['obs'], 'obs')
+    assert abs(z.mean()) < 0.1
+    back = norm.unnormalize(z, 'obs')
+    assert np.allclose(back, data['obs'], atol=1e-4)
+
+
+def test_cluster_script_generation(tmp_path):
+    k8s = K8sLauncher()
+    manifest = k8s.create_manifest('test-job', 'pkg.main', workers=2, output_path=str(tmp_path / 'job.yaml'))
+    assert 'parallelism: 2' in manifest
+    script = generate_slurm_script('job', 'python bench.py', nodes=2, output_path=str(tmp_path / 'job.sh'))
+    assert '#SBATCH --nodes=2' in script
