@@ -14,3 +14,4 @@ from .env_manager.subprocess_env_manager import (
     SyncSubprocessEnvManager, AsyncSubprocessEnvManager, SubprocessEnvManagerV2,
 )
 from .common.spaces import Discrete, Box, MultiDiscrete
+from .env_manager.env_supervisor import EnvSupervisor

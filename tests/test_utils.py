@@ -99,3 +99,18 @@ def test_seed():
     a = torch.rand(3)
     set_pkg_seed(42, use_cuda=False)
     assert torch.equal(a, torch.rand(3))
+
+
+def test_env_supervisor():
+    from ding.envs import EnvSupervisor
+    from ding.framework.supervisor import ChildType
+    from dizoo.classic_control.cartpole.envs.cartpole_env import CartPoleEnv
+    sup = EnvSupervisor(ChildType.THREAD, env_fn=[lambda: CartPoleEnv({}) for _ in range(2)])
+    sup.seed(0)
+    sup.launch()
+    obs = sup.ready_obs
+    assert len(obs) == 2
+    import numpy as np
+    for _ in range(5):
+        ts = sup.step({i: np.random.randint(2) for i in sup.ready_obs_id})
+    sup.close()
