@@ -10,3 +10,6 @@ from .decision_transformer import DecisionTransformer
 from .acer_model import ACER
 from .pdqn import PDQN
 from .qac_dist import QACDIST
+from .ebm import EBM, DFO, LangevinMCMC
+from .vae import VanillaVAE
+from .language_transformer import LanguageTransformer
