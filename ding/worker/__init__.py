@@ -3,3 +3,5 @@ from .collector.sample_serial_collector import SampleSerialCollector, EpisodeSer
 from .collector.interaction_serial_evaluator import InteractionSerialEvaluator, create_serial_evaluator
 from .coordinator.base_serial_commander import BaseSerialCommander
 from .replay_buffer.naive_buffer import NaiveReplayBuffer, AdvancedReplayBuffer, EpisodeReplayBuffer, create_buffer, get_buffer_cls, IBuffer
+from .coordinator.coordinator import Coordinator, ResourceManager
+from .adapter.learner_aggregator import LearnerAggregator
