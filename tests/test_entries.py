@@ -73,3 +73,40 @@ def test_dyna_pipeline():
 def test_cli_imports():
     from ding.entry import cli, cli_ditask
     assert callable(cli) and callable(cli_ditask)
+
+
+def test_atari_lite_pong_ppo_short():
+    """BASELINE config #2 plumbing on CPU: tiny n_sample, base manager."""
+    import copy
+    from ding.entry import serial_pipeline_onpolicy
+    from dizoo.atari.config.serial.pong_ppo_config import main_config, create_config
+    main = copy.deepcopy(main_config)
+    create = copy.deepcopy(create_config)
+    main.exp_name = 'exp/test_pong_ppo_lite'
+    main.policy.cuda = False
+    main.env.collector_env_num = 2
+    main.env.evaluator_env_num = 2
+    main.env.n_evaluator_episode = 2
+    main.policy.collect.n_sample = 32
+    main.policy.learn.batch_size = 16
+    main.policy.learn.epoch_per_collect = 1
+    create.env_manager.type = 'base'
+    serial_pipeline_onpolicy((main, create), seed=0, max_train_iter=2)
+
+
+def test_atari_lite_impala_short():
+    import copy
+    from ding.entry import serial_pipeline
+    from dizoo.atari.config.serial.spaceinvaders_impala_config import main_config, create_config
+    main = copy.deepcopy(main_config)
+    create = copy.deepcopy(create_config)
+    main.exp_name = 'exp/test_impala_lite'
+    main.policy.cuda = False
+    main.env.collector_env_num = 2
+    main.env.evaluator_env_num = 2
+    main.env.n_evaluator_episode = 2
+    main.policy.unroll_len = 8
+    main.policy.collect.n_sample = 4
+    main.policy.learn.batch_size = 2
+    create.env_manager.type = 'base'
+    serial_pipeline((main, create), seed=0, max_train_iter=2)
