@@ -115,10 +115,15 @@ class PPOPolicy(Policy):
             if self._recompute_adv:
                 with torch.no_grad():
                     if isinstance(data['obs'], torch.Tensor):
-                        # one batched critic pass over [obs; next_obs] — halves
-                        # the number of large forwards vs two separate calls
+                        # one batched critic pass over [obs; next_obs], chunked
+                        # to <=3200 rows: MIOpen's tuned conv solvers cover the
+                        # minibatch-family shapes; huge batches fall to naive
+                        # fp64-accum kernels (seen in rocprof on MI355X)
                         both = torch.cat([data['obs'], data['next_obs']], dim=0)
-                        values = self._learn_model.forward(both, mode='compute_critic')['value']
+                        chunks = torch.split(both, 3200, dim=0)
+                        values = torch.cat(
+                            [self._learn_model.forward(c, mode='compute_critic')['value'] for c in chunks], dim=0
+                        )
                         value, next_value = values.chunk(2, dim=0)
                     else:
                         value = self._learn_model.forward(data['obs'], mode='compute_critic')['value']

@@ -31,6 +31,8 @@ def default_collate(batch: Sequence, cat_1dim: bool = True, ignore_prefix: list 
         return torch.tensor(batch, dtype=torch.float32)
     if isinstance(elem, (np.integer, int)):
         return torch.tensor(batch)
+    if isinstance(elem, np.bool_):
+        return torch.tensor(batch)
     if isinstance(elem, (str, bytes)):
         return list(batch)
     if isinstance(elem, bool):
