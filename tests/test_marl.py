@@ -54,3 +54,16 @@ def test_mappo():
                  collect=dict(n_sample=32, unroll_len=1, discount_factor=0.99, gae_lambda=0.95))
     main, create = _marl_cfg('ppo', model, extra)
     serial_pipeline_onpolicy((main, create), seed=0, max_train_iter=2)
+
+
+def test_happo():
+    extra = dict(
+        agent_num=3,
+        action_space='discrete',
+        learn=dict(epoch_per_collect=1, batch_size=16, learning_rate=3e-4),
+        collect=dict(n_sample=32, unroll_len=1, discount_factor=0.99, gae_lambda=0.95),
+    )
+    model = dict(agent_obs_shape=8, global_obs_shape=12, action_shape=4, agent_num=3,
+                 actor_hidden_size_list=[32, 32], critic_hidden_size_list=[32, 32])
+    main, create = _marl_cfg('happo', model, extra)
+    serial_pipeline_onpolicy((main, create), seed=0, max_train_iter=2)
