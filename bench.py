@@ -25,10 +25,6 @@ import os
 import time
 
 import os
-# MIOpen: enforce tuned-solver search+db so every conv shape gets its best
-# kernel (A/B on MI355X: +10% over default find mode)
-os.environ.setdefault('MIOPEN_FIND_ENFORCE', '3')
-
 import torch
 
 # MIOpen autotune: pick the fastest conv algorithms for the fixed bench shapes

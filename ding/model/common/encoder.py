@@ -44,7 +44,7 @@ class UnfoldConv2d(nn.Module):
         Wo = (W + 2 * p - k) // s + 1
         cols = torch.nn.functional.unfold(x, k, stride=s, padding=p)  # [B, C*k*k, Ho*Wo]
         w = self.conv.weight.reshape(self.conv.out_channels, -1)  # [O, C*k*k]
-        out = torch.einsum('oc,bcl->bol', w, cols) + self.conv.bias.reshape(1, -1, 1)
+        out = torch.matmul(w, cols) + self.conv.bias.reshape(1, -1, 1)  # batched hipBLASLt GEMM
         return out.reshape(B, self.conv.out_channels, Ho, Wo)
 
 
@@ -69,7 +69,7 @@ class ConvEncoder(nn.Module):
             padding = [0] * len(kernel_size)
         if fast_im2col is None:
             import os
-            fast_im2col = os.environ.get('DING_FAST_IM2COL', '1') not in ('0', 'false')
+            fast_im2col = os.environ.get('DING_FAST_IM2COL', '0') in ('1', 'true')
         layers = []
         in_c = obs_shape[0]
         for i, (k, s, p) in enumerate(zip(kernel_size, stride, padding)):
