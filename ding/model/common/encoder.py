@@ -75,7 +75,6 @@ class ConvEncoder(nn.Module):
         for i, (k, s, p) in enumerate(zip(kernel_size, stride, padding)):
             if fast_im2col and k >= 5:
                 # large-kernel strided conv: batched unfold+GEMM (see UnfoldConv2d)
-                from ding.torch_utils import build_activation
                 layers.append(UnfoldConv2d(in_c, hidden_size_list[i], k, s, p))
                 layers.append(build_activation(activation))
             else:
