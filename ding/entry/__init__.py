@@ -8,3 +8,9 @@ from .application_entry import eval, collect_demo_data, collect_episodic_demo_da
 from .parallel_entry import parallel_pipeline
 from .cli import cli
 from .cli_ditask import cli_ditask
+from .serial_entry_variants2 import (
+    serial_pipeline_ngu, serial_pipeline_r2d3, serial_pipeline_preference_based_irl,
+    serial_pipeline_preference_based_irl_onpolicy, serial_pipeline_trex, serial_pipeline_trex_onpolicy,
+    serial_pipeline_guided_cost, serial_pipeline_td3_vae, serial_pipeline_onpolicy_ppg,
+    serial_pipeline_bco, serial_pipeline_pc, trex_collecting_data,
+)

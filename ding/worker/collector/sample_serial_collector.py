@@ -56,7 +56,10 @@ class SampleSerialCollector(ISerialCollector):
     def reset_env(self, _env: Optional[BaseEnvManager] = None) -> None:
         if _env is not None:
             self._env = _env
-            self._env.launch()
+            if self._env.closed:
+                self._env.launch()
+            else:
+                self._env.reset()
             self._env_num = self._env.env_num
         else:
             self._env.reset()

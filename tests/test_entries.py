@@ -110,3 +110,60 @@ def test_atari_lite_impala_short():
     main.policy.learn.batch_size = 2
     create.env_manager.type = 'base'
     serial_pipeline((main, create), seed=0, max_train_iter=2)
+
+
+def test_r2d3_pipeline():
+    from ding.entry import serial_pipeline_r2d3
+    main, create = cartpole_cfg('dqn')
+    main.policy.collect.pho = 0.25
+    main.policy.learn.expert_replay_buffer_size = 32
+    em, ec = cartpole_cfg('dqn')
+    serial_pipeline_r2d3((main, create), (em, ec), seed=0, max_train_iter=2)
+
+
+def test_ngu_entry_pipeline():
+    from ding.entry import serial_pipeline_ngu
+    main, create = cartpole_cfg('dqn')
+    main.rnd_reward_model = EasyDict(dict(
+        type='rnd-ngu', obs_shape=4, hidden_size_list=[16, 16], update_per_collect=1, batch_size=8
+    ))
+    main.episodic_reward_model = EasyDict(dict(
+        type='episodic', obs_shape=4, hidden_size_list=[16, 16], update_per_collect=1, batch_size=8
+    ))
+    serial_pipeline_ngu((main, create), seed=0, max_train_iter=2)
+
+
+def test_trex_pipeline(tmp_path):
+    import pickle as pkl
+    from ding.entry import serial_pipeline_preference_based_irl
+    main, create = cartpole_cfg('dqn')
+    # fabricate ranked demo episodes: obs sequences + returns
+    episodes = [[torch.randn(4) for _ in range(8)] for _ in range(4)]
+    returns = [1.0, 2.0, 3.0, 4.0]
+    with open(tmp_path / 'episodes_data.pkl', 'wb') as f:
+        pkl.dump(episodes, f)
+    with open(tmp_path / 'learning_returns.pkl', 'wb') as f:
+        pkl.dump(returns, f)
+    main.reward_model = EasyDict(dict(
+        type='trex', obs_shape=4, hidden_size_list=[16, 16], update_per_collect=2,
+        data_path=str(tmp_path), num_snippets=8, snippet_length=4,
+    ))
+    serial_pipeline_preference_based_irl((main, create), seed=0, max_train_iter=2)
+
+
+def test_onpolicy_ppg_pipeline():
+    from ding.entry import serial_pipeline_onpolicy_ppg
+    main, create = cartpole_cfg('ppg', extra_policy=dict(
+        learn=dict(epoch_per_collect=1, aux_freq=1),
+        collect=dict(discount_factor=0.99, gae_lambda=0.95),
+    ))
+    del main.policy.other  # on-policy: no replay buffer needed but _build_workers wants one
+    main.policy.other = EasyDict(dict(replay_buffer=dict(type='naive', replay_buffer_size=100)))
+    serial_pipeline_onpolicy_ppg((main, create), seed=0, max_train_iter=2)
+
+
+def test_bco_pipeline():
+    from ding.entry import serial_pipeline_bco
+    main, create = cartpole_cfg('bc')
+    em, ec = cartpole_cfg('dqn')
+    serial_pipeline_bco((main, create), (em, ec), seed=0, max_train_iter=2)
