@@ -36,6 +36,9 @@ class BaseLeague:
 
     def __init__(self, cfg: EasyDict):
         self.cfg = deep_merge_dicts(self.default_config(), cfg or EasyDict({}))
+        # active_players is a choice, not a merge target: user config replaces it
+        if cfg and 'active_players' in cfg:
+            self.cfg.active_players = EasyDict(cfg.active_players)
         self.path_policy = self.cfg.path_policy
         os.makedirs(self.path_policy, exist_ok=True)
         self.active_players: List[ActivePlayer] = []
@@ -57,11 +60,28 @@ class BaseLeague:
                     self.active_players.append(player)
                     self.payoff.add_player(player)
 
+    @property
+    def active_players_ids(self) -> List[str]:
+        return [p.player_id for p in self.active_players]
+
+    @property
+    def active_players_ckpts(self) -> List[str]:
+        return [p.checkpoint_path for p in self.active_players]
+
     # ------------------------------------------------------------- jobs
     def get_job_info(self, player_id: str, eval_flag: bool = False) -> dict:
         player = self.get_player_by_id(player_id)
         job = player.get_job(eval_flag)
         opponent = job['opponent']
+        if eval_flag:
+            return {
+                'agent_num': 1,
+                'launch_player': player_id,
+                'player_id': [player_id],
+                'checkpoint_path': [player.checkpoint_path],
+                'player_active_flag': [isinstance(player, ActivePlayer)],
+                'eval_opponent': opponent,
+            }
         return {
             'agent_num': 2,
             'launch_player': player_id,

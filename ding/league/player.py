@@ -150,6 +150,13 @@ class ActivePlayer(Player):
         return self._pfsp_branch()
 
 
+@PLAYER_REGISTRY.register('naive_sp_player')
+class NaiveSpPlayer(ActivePlayer):
+    """Simplest battle league player: PFSP vs snapshots when available,
+    plain self-play otherwise (reference ding/league/player.py NaiveSpPlayer)."""
+    _name = "NaiveSpPlayer"
+
+
 @PLAYER_REGISTRY.register('main_player')
 class MainPlayer(ActivePlayer):
     _name = "MainPlayer"

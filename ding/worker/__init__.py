@@ -5,3 +5,5 @@ from .coordinator.base_serial_commander import BaseSerialCommander
 from .replay_buffer.naive_buffer import NaiveReplayBuffer, AdvancedReplayBuffer, EpisodeReplayBuffer, create_buffer, get_buffer_cls, IBuffer
 from .coordinator.coordinator import Coordinator, ResourceManager
 from .adapter.learner_aggregator import LearnerAggregator
+from .collector.battle_collector import BattleSampleSerialCollector, BattleEpisodeSerialCollector
+from .collector.metric_serial_evaluator import MetricSerialEvaluator, IMetric
