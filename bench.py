@@ -64,6 +64,9 @@ def build_ppo_policy(device: str, multi_gpu: bool):
             ignore_done=False,
             grad_clip_type='clip_norm',
             grad_clip_value=0.5,
+            # hipGraph-capture the minibatch fwd+loss+bwd (launch-bound step;
+            # single-process only — the policy ignores this under multi_gpu)
+            cuda_graph=os.environ.get('DING_PPO_GRAPH', '1') not in ('0', 'false'),
         ),
         collect=dict(n_sample=3200, unroll_len=1, discount_factor=0.99, gae_lambda=0.95),
     )

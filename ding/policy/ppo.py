@@ -99,7 +99,47 @@ class PPOPolicy(Policy):
         self._gamma = self._cfg.collect.discount_factor
         self._gae_lambda = self._cfg.collect.gae_lambda
         self._recompute_adv = self._cfg.recompute_adv
+        # hipGraph capture of the minibatch step (MI355X: the step is
+        # launch-bound — see ding/torch_utils/hip_graph.py). Single-process
+        # only: the bucketed DDP reducer's backward hooks are not replayed
+        # by graphs, so multi_gpu keeps the eager path.
+        self._cuda_graph = self._cfg.learn.get('cuda_graph', False) and not self._cfg.multi_gpu
+        self._graphed_step = None
         self._learn_model.reset()
+
+    def _graphed_minibatch(self, batch: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
+        """Replay (or first capture) the fwd+loss+bwd hipGraph for one
+        minibatch; optimizer step stays eager (grad-clip Adam is not
+        capture-safe). Returns 0-dim GPU tensors — read them lazily."""
+        if self._graphed_step is None:
+            from ding.torch_utils.hip_graph import GraphedStep
+            from ding.ops import dispatch as _dispatch
+            wv, we = self._value_weight, self._entropy_weight
+
+            def step_fn(b):
+                output = self._learn_model.forward(b['obs'], mode='compute_actor_critic')
+                adv = b['adv']
+                if self._adv_norm:
+                    adv = (adv - adv.mean()) / (adv.std() + 1e-8)
+                policy_loss, value_loss, entropy_loss, approx_kl, clipfrac = _dispatch.fused_ppo_error(
+                    output['logit'], b['logit'], b['action'], output['value'], b['value'],
+                    adv, b['return'], None, self._clip_ratio, True
+                )
+                total_loss = policy_loss + wv * value_loss - we * entropy_loss
+                self._optimizer.zero_grad(set_to_none=False)
+                total_loss.backward()
+                return {
+                    'total_loss': total_loss.detach(), 'policy_loss': policy_loss.detach(),
+                    'value_loss': value_loss.detach(), 'entropy_loss': entropy_loss.detach(),
+                    'approx_kl': approx_kl, 'clipfrac': clipfrac,
+                    'adv_max': adv.max().detach(), 'adv_mean': adv.mean().detach(),
+                    'value_mean': output['value'].mean().detach(), 'value_max': output['value'].max().detach(),
+                }
+
+            self._graphed_step = GraphedStep(step_fn)
+        out = self._graphed_step(batch)
+        self._optimizer.step()
+        return out
 
     def _forward_learn(self, data: List[Dict[str, Any]]) -> List[Dict[str, Any]]:
         data = default_preprocess_learn(data, ignore_done=self._cfg.learn.ignore_done, use_nstep=False)
@@ -150,7 +190,18 @@ class PPOPolicy(Policy):
                 else:
                     data['return'] = data['adv'] + data['value']
 
+            graph_infos = []
             for batch in split_data_generator(data, self._cfg.learn.batch_size, shuffle=True):
+                if (
+                    self._cuda_graph and self._action_space == 'discrete'
+                    and isinstance(batch['obs'], torch.Tensor) and batch['obs'].is_cuda
+                    and batch.get('weight') is None and batch['obs'].dtype == torch.float32
+                ):
+                    out = self._graphed_minibatch(batch)
+                    # static outputs: clone (async) now, convert to floats once
+                    # at the end of the epoch — avoids a device sync per minibatch
+                    graph_infos.append({k: v.clone() for k, v in out.items()})
+                    continue
                 output = self._learn_model.forward(batch['obs'], mode='compute_actor_critic')
                 adv = batch['adv']
                 if self._adv_norm:
@@ -214,6 +265,13 @@ class PPOPolicy(Policy):
                     'clipfrac': ppo_info.clipfrac,
                 }
                 return_infos.append(return_info)
+            if graph_infos:
+                # one host sync for the whole epoch's graphed minibatches
+                lr = self._optimizer.defaults['lr']
+                for g in graph_infos:
+                    info = {k: float(v) for k, v in g.items()}
+                    info['cur_lr'] = lr
+                    return_infos.append(info)
         return return_infos
 
     def _monitor_vars_learn(self) -> List[str]:
