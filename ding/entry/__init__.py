@@ -2,7 +2,7 @@ from .serial_entry import serial_pipeline, serial_pipeline_onpolicy, random_coll
 from .serial_entry_offline import serial_pipeline_offline
 from .serial_entry_variants import (
     serial_pipeline_reward_model, serial_pipeline_sqil, serial_pipeline_gail, serial_pipeline_dqfd,
-    serial_pipeline_bc, serial_pipeline_dyna,
+    serial_pipeline_bc, serial_pipeline_dyna, serial_pipeline_dream,
 )
 from .application_entry import eval, collect_demo_data, collect_episodic_demo_data, episode_to_transitions
 from .parallel_entry import parallel_pipeline

@@ -317,3 +317,46 @@ def serial_pipeline_dyna(
     evaluator.close()
     learner.close()
     return policy
+
+
+def serial_pipeline_dream(
+    input_cfg: Union[str, Tuple[dict, dict]],
+    seed: int = 0,
+    env_setting: Optional[List[Any]] = None,
+    model: Optional[torch.nn.Module] = None,
+    max_train_iter: int = int(1e10),
+    max_env_step: int = int(1e10),
+) -> 'Policy':  # noqa
+    """Dream-style MBRL (MBSAC/STEVESAC): the policy trains on differentiable
+    imagined rollouts, so the learner forwards the world model + envstep into
+    policy._forward_learn (reference serial_entry_mbrl.py:176)."""
+    from ding.world_model import create_world_model
+    cfg, policy, collector_env, evaluator_env = _prepare(input_cfg, seed, env_setting, model)
+    learner, collector, evaluator, replay_buffer, commander = _build_workers(
+        cfg, policy, collector_env, evaluator_env
+    )
+    world_model = create_world_model(cfg.world_model, env=None)
+    learner.call_hook('before_run')
+    while True:
+        collect_kwargs = commander.step()
+        if evaluator.should_eval(learner.train_iter):
+            stop, _ = evaluator.eval(learner.save_checkpoint, learner.train_iter, collector.envstep)
+            if stop:
+                break
+        new_data = collector.collect(train_iter=learner.train_iter, policy_kwargs=collect_kwargs)
+        replay_buffer.push(new_data, cur_collector_envstep=collector.envstep)
+        if world_model.should_train(collector.envstep):
+            world_model.train(replay_buffer, collector.envstep, learner.train_iter)
+        upc = max(1, cfg.policy.learn.update_per_collect // max(1, world_model.rollout_length_scheduler(collector.envstep)))
+        for _ in range(upc):
+            train_data = replay_buffer.sample(learner.policy.get_attribute('batch_size'), learner.train_iter)
+            if train_data is None:
+                break
+            learner.train(
+                train_data, collector.envstep,
+                policy_kwargs=dict(world_model=world_model, envstep=collector.envstep)
+            )
+        if collector.envstep >= max_env_step or learner.train_iter >= max_train_iter:
+            break
+    learner.call_hook('after_run')
+    return policy

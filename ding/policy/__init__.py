@@ -19,3 +19,4 @@ from .r2d2_variants import NGUPolicy, R2D3Policy, R2D2GTrXLPolicy, R2D2CollectTr
 from .misc_policies import IBCPolicy, BCQPolicy, TD3VAEPolicy, PromptPGPolicy, PromptAWRPolicy, ProcedureCloningBFSPolicy
 from .happo import HAPPOPolicy
 from . import command_mode_policy_instance  # registers '<name>_command' variants
+from .mbpolicy import MBSACPolicy, STEVESACPolicy

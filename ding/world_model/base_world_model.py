@@ -145,18 +145,37 @@ class DreamWorldModel(WorldModel, ABC):
     (reference base_world_model.py:269 rollout)."""
 
     def rollout(self, obs: torch.Tensor, actor_fn: Callable, envstep: int,
-                **kwargs) -> Tuple[Any, ...]:
+                **kwargs) -> Tuple[torch.Tensor, ...]:
+        """Differentiable imagination rollout for BPTT value gradients.
+
+        Returns stacked tensors (reference base_world_model.py rollout):
+        obss [N+1,B,O] (obss[0] real), actions [N+1,B,A] (incl. bootstrap
+        action at the final state), rewards [N,B] (entropy-augmented),
+        aug_rewards [N+1,B], dones [N,B]. Gradients flow to the POLICY only —
+        the world model is frozen for the duration of the rollout."""
+        import torch.nn as nn
         horizon = self.rollout_length_scheduler(envstep)
+        if isinstance(self, nn.Module):
+            self.requires_grad_(False)
         obss, actions, rewards, aug_rewards, dones = [obs], [], [], [], []
         for _ in range(horizon):
             action, aug_reward = actor_fn(obs)
-            reward, obs, done = self.step(obs, action)
+            reward, obs, done = self.step(obs, action, **kwargs)
+            reward = reward + aug_reward
             obss.append(obs)
             actions.append(action)
             rewards.append(reward)
             aug_rewards.append(aug_reward)
             dones.append(done)
-        return obss, actions, rewards, aug_rewards, dones
+        action, aug_reward = actor_fn(obs)
+        actions.append(action)
+        aug_rewards.append(aug_reward)
+        if isinstance(self, nn.Module):
+            self.requires_grad_(True)
+        return (
+            torch.stack(obss), torch.stack(actions), torch.stack(rewards),
+            torch.stack(aug_rewards), torch.stack(dones)
+        )
 
 
 class HybridWorldModel(DreamWorldModel, DynaWorldModel, ABC):
