@@ -529,3 +529,55 @@ def trex_collecting_data(args=None) -> None:
         pickle.dump(episodes, f)
     with open(os.path.join(data_path, 'learning_returns.pkl'), 'wb') as f:
         pickle.dump(returns, f)
+
+
+def serial_pipeline_dreamer(
+    input_cfg: Union[str, Tuple[dict, dict]],
+    seed: int = 0,
+    env_setting: Optional[List[Any]] = None,
+    model: Optional[torch.nn.Module] = None,
+    max_train_iter: int = int(1e10),
+    max_env_step: int = int(1e10),
+) -> 'Policy':  # noqa
+    """DreamerV3: RSSM world model trained on sequence batches; policy trained
+    purely in latent imagination from the posterior; collect/eval thread the
+    RSSM filter state (reference serial_entry_mbrl.py:249)."""
+    from ding.world_model import create_world_model
+    cfg, policy, collector_env, evaluator_env = _prepare(input_cfg, seed, env_setting, model)
+    learner, collector, evaluator, env_buffer, commander = _build_workers(
+        cfg, policy, collector_env, evaluator_env
+    )
+    world_model = create_world_model(cfg.world_model, env=None)
+    learner.call_hook('before_run')
+    from .serial_entry import random_collect as _random_collect
+    if cfg.policy.get('random_collect_size', 0) > 0:
+        _random_collect(cfg.policy, policy, collector, collector_env, commander, env_buffer)
+    while True:
+        collect_kwargs = commander.step()
+        if evaluator.should_eval(learner.train_iter):
+            stop, _ = evaluator.eval(
+                learner.save_checkpoint, learner.train_iter, collector.envstep,
+                policy_kwargs=dict(world_model=world_model)
+            )
+            if stop:
+                break
+        steps = cfg.world_model.get('pretrain', 1) if world_model.should_pretrain() \
+            else int(world_model.should_train(collector.envstep))
+        for _ in range(steps):
+            post, context = world_model.train(
+                env_buffer, collector.envstep, learner.train_iter,
+                learner.policy.get_attribute('batch_size'), cfg.policy.learn.batch_length
+            )
+            learner.train(
+                post, collector.envstep,
+                policy_kwargs=dict(world_model=world_model, envstep=collector.envstep)
+            )
+        data = collector.collect(
+            train_iter=learner.train_iter,
+            policy_kwargs=dict(world_model=world_model, envstep=collector.envstep)
+        )
+        env_buffer.push(data, cur_collector_envstep=collector.envstep)
+        if collector.envstep >= max_env_step or learner.train_iter >= max_train_iter:
+            break
+    learner.call_hook('after_run')
+    return policy

@@ -160,3 +160,46 @@ class VAC(nn.Module):
         if 'unnormalized_pred' in out:
             ret['unnormalized_value'] = out['unnormalized_pred'].squeeze(-1)
         return ret
+
+
+@MODEL_REGISTRY.register('dreamervac')
+class DREAMERVAC(nn.Module):
+    """DreamerV3 actor-critic over RSSM latent features: ActionHead actor
+    (unimix one-hot / trunc-normal) + two-hot-symlog critic.
+
+    Parity: reference ding/model/template/vac.py DREAMERVAC:387.
+    """
+    mode = ['compute_actor', 'compute_critic', 'compute_actor_critic']
+
+    def __init__(
+        self,
+        action_shape,
+        dyn_stoch: int = 32,
+        dyn_deter: int = 512,
+        dyn_discrete: int = 32,
+        actor_layers: int = 2,
+        value_layers: int = 2,
+        units: int = 512,
+        act: str = 'SiLU',
+        norm: str = 'LN',
+        actor_dist: str = 'onehot',
+        actor_init_std: float = 1.0,
+        actor_min_std: float = 0.1,
+        actor_max_std: float = 1.0,
+        actor_temp: float = 0.1,
+        action_unimix_ratio: float = 0.01,
+        **kwargs,
+    ) -> None:
+        super().__init__()
+        from ding.torch_utils.network.dreamer import ActionHead, DenseHead
+        action_shape = squeeze(action_shape)
+        self.action_shape = action_shape
+        feat_size = dyn_stoch * dyn_discrete + dyn_deter if dyn_discrete else dyn_stoch + dyn_deter
+        self.actor = ActionHead(
+            feat_size, action_shape, actor_layers, units, act, norm, actor_dist,
+            actor_init_std, actor_min_std, actor_max_std, actor_temp,
+            outscale=1.0, unimix_ratio=action_unimix_ratio,
+        )
+        self.critic = DenseHead(
+            feat_size, (255, ), value_layers, units, act, norm, dist='twohot_symlog', outscale=0.0
+        )

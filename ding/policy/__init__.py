@@ -20,3 +20,7 @@ from .misc_policies import IBCPolicy, BCQPolicy, TD3VAEPolicy, PromptPGPolicy, P
 from .happo import HAPPOPolicy
 from . import command_mode_policy_instance  # registers '<name>_command' variants
 from .mbpolicy import MBSACPolicy, STEVESACPolicy
+from .dreamer import DREAMERPolicy
+# late-registered policies (dreamer/mbpolicy) also need '_command' variants
+from .command_mode_policy_instance import _register_command_variants as _rcv
+_rcv()

@@ -1,0 +1,527 @@
+"""DreamerV3 building blocks: distributions (symlog two-hot, unimix one-hot,
+truncated normal), heads, layer-norm GRU and the static_scan unroll helper.
+
+Parity: reference ding/torch_utils/network/dreamer.py (DenseHead:92,
+ActionHead:185, SampleDist:315, OneHotDist:365, TwoHotDistSymlog:419,
+SymlogDist:504, ContDist:571, Bernoulli:631, UnnormalizedHuber:702,
+SafeTruncatedNormal:741, TanhBijector:785, static_scan:838, weight_init).
+Re-designed: plain-Python math, fp32-first, no device strings threaded
+through constructors (modules follow `.to(device)` like the rest of ding).
+"""
+import math
+from typing import Callable, List, Optional, Tuple
+
+import numpy as np
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+import torch.distributions as torchd
+
+
+def symlog(x: torch.Tensor) -> torch.Tensor:
+    return torch.sign(x) * torch.log(1 + torch.abs(x))
+
+
+def symexp(x: torch.Tensor) -> torch.Tensor:
+    return torch.sign(x) * (torch.exp(torch.abs(x)) - 1)
+
+
+class Conv2dSame(nn.Conv2d):
+    """TF-style 'same' padding conv (odd/even input sizes both handled)."""
+
+    def _same_pad(self, i: int, k: int, s: int, d: int) -> int:
+        return max((math.ceil(i / s) - 1) * s + (k - 1) * d + 1 - i, 0)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        ih, iw = x.shape[-2:]
+        pad_h = self._same_pad(ih, self.kernel_size[0], self.stride[0], self.dilation[0])
+        pad_w = self._same_pad(iw, self.kernel_size[1], self.stride[1], self.dilation[1])
+        if pad_h > 0 or pad_w > 0:
+            x = F.pad(x, [pad_w // 2, pad_w - pad_w // 2, pad_h // 2, pad_h - pad_h // 2])
+        return F.conv2d(x, self.weight, self.bias, self.stride, self.padding, self.dilation, self.groups)
+
+
+class DreamerLayerNorm(nn.Module):
+    """Channel-wise LN for NCHW feature maps."""
+
+    def __init__(self, ch: int, eps: float = 1e-3):
+        super().__init__()
+        self.norm = nn.LayerNorm(ch, eps=eps)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.norm(x.permute(0, 2, 3, 1)).permute(0, 3, 1, 2)
+
+
+# --------------------------------------------------------------------------
+# distributions
+# --------------------------------------------------------------------------
+
+
+class SampleDist:
+    """Empirical mean/mode/entropy of a base distribution via sampling."""
+
+    def __init__(self, dist, samples: int = 100):
+        self._dist = dist
+        self._samples = samples
+
+    def __getattr__(self, name):
+        return getattr(self._dist, name)
+
+    def mean(self):
+        return self._dist.rsample((self._samples, )).mean(0)
+
+    def mode(self):
+        s = self._dist.rsample((self._samples, ))
+        logp = self._dist.log_prob(s)
+        return s[logp.argmax()][None].squeeze(0)
+
+    def entropy(self):
+        s = self._dist.rsample((self._samples, ))
+        return -self._dist.log_prob(s).mean(0)
+
+
+class OneHotDist(torchd.one_hot_categorical.OneHotCategorical):
+    """One-hot categorical with uniform mixing (DreamerV3 unimix) and a
+    straight-through rsample."""
+
+    def __init__(self, logits=None, probs=None, unimix_ratio: float = 0.0):
+        if logits is not None and unimix_ratio > 0.0:
+            probs = F.softmax(logits, dim=-1)
+            probs = probs * (1.0 - unimix_ratio) + unimix_ratio / probs.shape[-1]
+            logits = torch.log(probs)
+            super().__init__(logits=logits, probs=None)
+        else:
+            super().__init__(logits=logits, probs=probs)
+
+    def mode(self):
+        _mode = F.one_hot(torch.argmax(super().logits, axis=-1), super().logits.shape[-1])
+        return _mode.detach() + super().logits - super().logits.detach()
+
+    def sample(self, sample_shape=(), seed=None):
+        sample = super().sample(sample_shape)
+        probs = super().probs
+        while len(probs.shape) < len(sample.shape):
+            probs = probs[None]
+        return sample + (probs - probs.detach())  # straight-through gradients
+
+    rsample = sample
+
+
+class TwoHotDistSymlog:
+    """255-bin two-hot distribution over symlog-transformed scalars
+    (DreamerV3 reward/value heads)."""
+
+    def __init__(self, logits: torch.Tensor, low: float = -20.0, high: float = 20.0):
+        self.logits = logits
+        self.probs = torch.softmax(logits, -1)
+        self.buckets = torch.linspace(low, high, steps=255, device=logits.device)
+        self.width = (high - low) / 255
+
+    def mean(self):
+        _mean = self.probs * self.buckets
+        return symexp(torch.sum(_mean, dim=-1, keepdim=True))
+
+    def mode(self):
+        return self.mean()
+
+    def log_prob(self, x: torch.Tensor) -> torch.Tensor:
+        x = symlog(x)
+        below = (self.buckets <= x[..., None]).to(torch.int32).sum(dim=-1) - 1
+        above = len(self.buckets) - (self.buckets > x[..., None]).to(torch.int32).sum(dim=-1)
+        below = torch.clip(below, 0, len(self.buckets) - 1)
+        above = torch.clip(above, 0, len(self.buckets) - 1)
+        equal = below == above
+        dist_to_below = torch.where(equal, 1, torch.abs(self.buckets[below] - x))
+        dist_to_above = torch.where(equal, 1, torch.abs(self.buckets[above] - x))
+        total = dist_to_below + dist_to_above
+        weight_below = dist_to_above / total
+        weight_above = dist_to_below / total
+        target = (
+            F.one_hot(below, num_classes=len(self.buckets)) * weight_below[..., None] +
+            F.one_hot(above, num_classes=len(self.buckets)) * weight_above[..., None]
+        )
+        log_pred = self.logits - torch.logsumexp(self.logits, -1, keepdim=True)
+        target = target.squeeze(-2)
+        return (target * log_pred).sum(-1)
+
+    def log_prob_target(self, target: torch.Tensor) -> torch.Tensor:
+        log_pred = self.logits - torch.logsumexp(self.logits, -1, keepdim=True)
+        return (target * log_pred).sum(-1)
+
+
+class SymlogDist:
+    """MSE/abs distance in symlog space presented as a distribution."""
+
+    def __init__(self, mode: torch.Tensor, dist: str = 'mse', aggregation: str = 'sum',
+                 dim_to_reduce: List[int] = [-1, -2, -3]):
+        self._mode = mode
+        self._dist = dist
+        self._agg = aggregation
+        self._dims = tuple(dim_to_reduce)
+
+    def mode(self):
+        return symexp(self._mode)
+
+    def mean(self):
+        return symexp(self._mode)
+
+    def log_prob(self, value: torch.Tensor) -> torch.Tensor:
+        assert self._mode.shape == value.shape
+        if self._dist == 'mse':
+            distance = (self._mode - symlog(value)) ** 2.0
+        elif self._dist == 'abs':
+            distance = torch.abs(self._mode - symlog(value))
+        else:
+            raise NotImplementedError(self._dist)
+        if self._agg == 'mean':
+            loss = distance.mean(self._dims)
+        else:
+            loss = distance.sum(self._dims)
+        return -loss
+
+
+class ContDist:
+    """Wrapper adding mode()/entropy() passthrough to an Independent dist."""
+
+    def __init__(self, dist=None):
+        super().__init__()
+        self._dist = dist
+        self.mean = dist.mean
+
+    def __getattr__(self, name):
+        return getattr(self._dist, name)
+
+    def entropy(self):
+        return self._dist.entropy()
+
+    def mode(self):
+        return self._dist.mean
+
+    def sample(self, sample_shape=()):
+        return self._dist.rsample(sample_shape)
+
+    def log_prob(self, x):
+        return self._dist.log_prob(x)
+
+
+class Bernoulli:
+    """Bernoulli with straight-through-free log_prob used by discount heads."""
+
+    def __init__(self, dist=None):
+        super().__init__()
+        self._dist = dist
+        self.mean = dist.mean
+
+    def __getattr__(self, name):
+        return getattr(self._dist, name)
+
+    def entropy(self):
+        return self._dist.entropy()
+
+    def mode(self):
+        return (self._dist.mean > 0.5).to(self._dist.mean.dtype)
+
+    def sample(self, sample_shape=()):
+        return self._dist.rsample(sample_shape)
+
+    def log_prob(self, x):
+        _logits = self._dist.base_dist.logits
+        log_probs0 = -F.softplus(_logits)
+        log_probs1 = -F.softplus(-_logits)
+        return (log_probs0 * (1 - x) + log_probs1 * x).squeeze(-1)
+
+
+class UnnormalizedHuber(torchd.normal.Normal):
+
+    def __init__(self, loc, scale, threshold: float = 1.0, **kwargs):
+        super().__init__(loc, scale, **kwargs)
+        self._threshold = threshold
+
+    def log_prob(self, event):
+        return -(torch.sqrt((event - self.mean) ** 2 + self._threshold ** 2) - self._threshold)
+
+    def mode(self):
+        return self.mean
+
+
+class SafeTruncatedNormal(torchd.normal.Normal):
+    """Normal clipped to [low, high] with gradient-preserving clip."""
+
+    def __init__(self, loc, scale, low, high, clip: float = 1e-6, mult: float = 1.0):
+        super().__init__(loc, scale)
+        self._low, self._high = low, high
+        self._clip, self._mult = clip, mult
+
+    def sample(self, sample_shape):
+        event = super().rsample(sample_shape)
+        if self._clip:
+            clipped = torch.clip(event, self._low + self._clip, self._high - self._clip)
+            event = event - event.detach() + clipped.detach()
+        if self._mult:
+            event *= self._mult
+        return event
+
+
+class TanhBijector(torchd.Transform):
+
+    def __init__(self, validate_args=False, name='tanh'):
+        super().__init__()
+        self.bijective = True
+        self.domain = torchd.constraints.real
+        self.codomain = torchd.constraints.interval(-1.0, 1.0)
+        self.name = name
+
+    def _call(self, x):
+        return torch.tanh(x)
+
+    def _inverse(self, y):
+        y = torch.where((torch.abs(y) <= 1.), torch.clamp(y, -0.99999997, 0.99999997), y)
+        return torch.atanh(y)
+
+    def log_abs_det_jacobian(self, x, y):
+        log2 = torch.math.log(2.0) if hasattr(torch, 'math') else math.log(2.0)
+        return 2.0 * (log2 - x - F.softplus(-2.0 * x))
+
+
+# --------------------------------------------------------------------------
+# heads
+# --------------------------------------------------------------------------
+
+
+def _mlp(inp_dim: int, units: int, layers: int, act: str = 'SiLU', norm: str = 'LN') -> Tuple[nn.Sequential, int]:
+    act_cls = getattr(nn, act) if isinstance(act, str) else act
+    mods = []
+    d = inp_dim
+    for _ in range(layers):
+        mods.append(nn.Linear(d, units, bias=False))
+        if norm == 'LN' or norm is nn.LayerNorm:
+            mods.append(nn.LayerNorm(units, eps=1e-3))
+        mods.append(act_cls())
+        d = units
+    return nn.Sequential(*mods), d
+
+
+class DenseHead(nn.Module):
+    """MLP head producing a distribution over `shape`:
+    dist in {normal, huber, binary, twohot_symlog, mse}."""
+
+    def __init__(
+        self,
+        inp_dim: int,
+        shape: Tuple,
+        layer_num: int,
+        units: int,
+        act: str = 'SiLU',
+        norm: str = 'LN',
+        dist: str = 'normal',
+        std: float = 1.0,
+        outscale: float = 1.0,
+        device: str = 'cpu',
+    ):
+        super().__init__()
+        self._shape = (shape, ) if isinstance(shape, int) else tuple(shape)
+        if len(self._shape) == 0:
+            self._shape = (1, )
+        self._dist = dist
+        self._std = std
+        self.mlp, d = _mlp(inp_dim, units, layer_num, act, norm)
+        out_units = 255 if dist == 'twohot_symlog' else int(np.prod(self._shape))
+        self.mean_layer = nn.Linear(d, out_units)
+        uniform_weight_init(outscale)(self.mean_layer)
+        if self._dist in ('normal', 'huber'):
+            self.std_layer = nn.Linear(d, int(np.prod(self._shape)))
+            uniform_weight_init(outscale)(self.std_layer)
+
+    def forward(self, features: torch.Tensor):
+        x = self.mlp(features)
+        mean = self.mean_layer(x)
+        if self._dist == 'normal':
+            std = 2 * torch.sigmoid(self.std_layer(x) / 2) + 0.1
+            return ContDist(torchd.independent.Independent(torchd.normal.Normal(mean, std), 1))
+        if self._dist == 'huber':
+            std = 2 * torch.sigmoid(self.std_layer(x) / 2) + 0.1
+            return ContDist(torchd.independent.Independent(UnnormalizedHuber(mean, std, 1.0), len(self._shape)))
+        if self._dist == 'binary':
+            return Bernoulli(torchd.independent.Independent(torchd.bernoulli.Bernoulli(logits=mean), len(self._shape)))
+        if self._dist == 'twohot_symlog':
+            return TwoHotDistSymlog(logits=mean)
+        if self._dist == 'mse':
+            return SymlogDist(mean, 'mse', 'sum', dim_to_reduce=[-1])
+        raise NotImplementedError(self._dist)
+
+
+class ActionHead(nn.Module):
+    """Actor head: one-hot (discrete, unimix) or trunc-normal/tanh-normal
+    (continuous) action distribution over imagined features."""
+
+    def __init__(
+        self,
+        inp_dim: int,
+        size: int,
+        layers: int,
+        units: int,
+        act: str = 'SiLU',
+        norm: str = 'LN',
+        dist: str = 'trunc_normal',
+        init_std: float = 0.0,
+        min_std: float = 0.1,
+        max_std: float = 1.0,
+        temp: float = 0.1,
+        outscale: float = 1.0,
+        unimix_ratio: float = 0.01,
+    ):
+        super().__init__()
+        self._size = size
+        self._dist = dist
+        self._min_std = min_std
+        self._max_std = max_std
+        self._init_std = init_std
+        self._temp = temp
+        self._unimix_ratio = unimix_ratio
+        self.mlp, d = _mlp(inp_dim, units, layers, act, norm)
+        out_units = size if dist in ('onehot', 'onehot_gumble') else 2 * size
+        self.dist_layer = nn.Linear(d, out_units)
+        uniform_weight_init(outscale)(self.dist_layer)
+
+    def forward(self, features: torch.Tensor):
+        x = self.mlp(features)
+        out = self.dist_layer(x)
+        if self._dist == 'onehot':
+            return OneHotDist(out, unimix_ratio=self._unimix_ratio)
+        if self._dist == 'onehot_gumble':
+            return ContDist(torchd.gumbel.Gumbel(out, 1 / self._temp))
+        mean, std = torch.split(out, self._size, -1)
+        if self._dist == 'trunc_normal':
+            mean = torch.tanh(mean)
+            std = 2 * torch.sigmoid(std / 2) + self._min_std
+            dist = SafeTruncatedNormal(mean, std, -1, 1)
+            return ContDist(torchd.independent.Independent(dist, 1))
+        if self._dist == 'normal':
+            std = (self._max_std - self._min_std) * torch.sigmoid(std + 2.0) + self._min_std
+            dist = torchd.normal.Normal(torch.tanh(mean), std)
+            return ContDist(torchd.independent.Independent(dist, 1))
+        if self._dist == 'tanh_normal':
+            mean = 5 * torch.tanh(mean / 5)
+            std = F.softplus(std + self._init_std) + self._min_std
+            dist = torchd.normal.Normal(mean, std)
+            dist = torchd.transformed_distribution.TransformedDistribution(dist, TanhBijector())
+            return SampleDist(torchd.independent.Independent(dist, 1))
+        raise NotImplementedError(self._dist)
+
+
+# --------------------------------------------------------------------------
+# unroll / init helpers
+# --------------------------------------------------------------------------
+
+
+def static_scan(fn: Callable, inputs: Tuple, start, reverse: bool = False):
+    """Unrolled scan: apply `fn(carry, *inputs_t)` along dim 0, stacking every
+    output. `start` may be a (nested) tuple of dict/tensor carries."""
+    last = start
+    flag = True
+    indices = range(inputs[0].shape[0])
+    outputs = None
+    for index in indices:
+        inp = (lambda x: (_inp[x] for _inp in inputs))(index)
+        last = fn(last, *inp)
+        if flag:
+            if isinstance(last, dict):
+                outputs = {key: value.clone().unsqueeze(0) for key, value in last.items()}
+            else:
+                outputs = []
+                for _last in last:
+                    if isinstance(_last, dict):
+                        outputs.append({key: value.clone().unsqueeze(0) for key, value in _last.items()})
+                    else:
+                        outputs.append(_last.clone().unsqueeze(0))
+            flag = False
+        else:
+            if isinstance(last, dict):
+                for key in last.keys():
+                    outputs[key] = torch.cat([outputs[key], last[key].unsqueeze(0)], dim=0)
+            else:
+                for j in range(len(outputs)):
+                    if isinstance(last[j], dict):
+                        for key in last[j].keys():
+                            outputs[j][key] = torch.cat([outputs[j][key], last[j][key].unsqueeze(0)], dim=0)
+                    else:
+                        outputs[j] = torch.cat([outputs[j], last[j].unsqueeze(0)], dim=0)
+    if isinstance(outputs, dict):
+        outputs = [outputs]
+    return outputs
+
+
+class GRUCellLN(nn.Module):
+    """GRU cell with layer-norm on the joint input-state projection and
+    update-gate bias (DreamerV3 'deter' cell)."""
+
+    def __init__(self, inp_size: int, size: int, norm: bool = False, act=torch.tanh, update_bias: float = -1):
+        super().__init__()
+        self._size = size
+        self._act = act
+        self._update_bias = update_bias
+        self.layers = nn.Sequential()
+        self.layers.add_module('linear', nn.Linear(inp_size + size, 3 * size, bias=False))
+        if norm:
+            self.layers.add_module('norm', nn.LayerNorm(3 * size, eps=1e-3))
+
+    @property
+    def state_size(self):
+        return self._size
+
+    def forward(self, inputs: torch.Tensor, state: List[torch.Tensor]):
+        state = state[0]
+        parts = self.layers(torch.cat([inputs, state], -1))
+        reset, cand, update = torch.split(parts, [self._size] * 3, -1)
+        reset = torch.sigmoid(reset)
+        cand = self._act(reset * cand)
+        update = torch.sigmoid(update + self._update_bias)
+        output = update * cand + (1 - update) * state
+        return output, [output]
+
+
+def weight_init(m):
+    """He-style init with layer-norm reset (DreamerV3 default)."""
+    if isinstance(m, nn.Linear):
+        in_num = m.in_features
+        out_num = m.out_features
+        denoms = (in_num + out_num) / 2.0
+        scale = 1.0 / denoms
+        std = np.sqrt(scale) / 0.87962566103423978
+        nn.init.trunc_normal_(m.weight.data, mean=0.0, std=std, a=-2.0 * std, b=2.0 * std)
+        if hasattr(m.bias, 'data'):
+            m.bias.data.fill_(0.0)
+    elif isinstance(m, (nn.Conv2d, nn.ConvTranspose2d)):
+        space = m.kernel_size[0] * m.kernel_size[1]
+        in_num = space * m.in_channels
+        out_num = space * m.out_channels
+        denoms = (in_num + out_num) / 2.0
+        scale = 1.0 / denoms
+        std = np.sqrt(scale) / 0.87962566103423978
+        nn.init.trunc_normal_(m.weight.data, mean=0.0, std=std, a=-2.0, b=2.0)
+        if hasattr(m.bias, 'data'):
+            m.bias.data.fill_(0.0)
+    elif isinstance(m, nn.LayerNorm):
+        m.weight.data.fill_(1.0)
+        if hasattr(m.bias, 'data'):
+            m.bias.data.fill_(0.0)
+
+
+def uniform_weight_init(given_scale: float):
+
+    def f(m):
+        if isinstance(m, nn.Linear):
+            in_num = m.in_features
+            out_num = m.out_features
+            denoms = (in_num + out_num) / 2.0
+            scale = given_scale / denoms
+            limit = np.sqrt(3 * scale)
+            nn.init.uniform_(m.weight.data, a=-limit, b=limit)
+            if hasattr(m.bias, 'data'):
+                m.bias.data.fill_(0.0)
+        elif isinstance(m, nn.LayerNorm):
+            m.weight.data.fill_(1.0)
+            if hasattr(m.bias, 'data'):
+                m.bias.data.fill_(0.0)
+
+    return f

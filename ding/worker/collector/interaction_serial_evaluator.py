@@ -117,18 +117,26 @@ class InteractionSerialEvaluator:
         self._env.reset()
         self._policy.reset()
         monitor = VectorEvalMonitor(self._env_num, n_episode)
+        # recurrent-filter policies (dreamer): thread latent state + resets
+        _recurrent = str(getattr(self._policy.get_attribute('cfg'), 'type', '')).startswith('dreamer')
+        _states, _resets = None, np.zeros(self._env_num)
         while not monitor.is_finished():
             obs = self._env.ready_obs
             if not isinstance(obs, dict) or not all(isinstance(k, int) for k in obs):
                 ids = self._env.ready_obs_id
                 obs = {i: obs[pos] for pos, i in enumerate(ids)}
             obs_t = {i: to_tensor(o, dtype=torch.float32) for i, o in obs.items()}
-            policy_output = self._policy.forward(obs_t, **(policy_kwargs or {}))
+            if _recurrent:
+                policy_output = self._policy.forward(obs_t, **(policy_kwargs or {}), reset=_resets, state=_states)
+                _states = [policy_output[i]['state'] for i in sorted(policy_output)]
+            else:
+                policy_output = self._policy.forward(obs_t, **(policy_kwargs or {}))
             actions = {i: to_ndarray(out['action']) for i, out in policy_output.items()}
             timesteps = self._env.step(actions)
             if not isinstance(timesteps, dict):
                 timesteps = {ts.info['env_id']: ts for ts in timesteps}
             for env_id, timestep in timesteps.items():
+                _resets[env_id] = float(bool(timestep.done))
                 if timestep.done:
                     self._policy.reset([env_id])
                     monitor.update_reward(env_id, timestep.info.get('eval_episode_return', 0.0))

@@ -81,6 +81,9 @@ class SampleSerialCollector(ISerialCollector):
         self._obs_pool = {}
         self._policy_output_pool = {}
         self._traj_buffer = {env_id: [] for env_id in range(self._env_num)}
+        # recurrent-filter policies (dreamer): per-env latent state + reset flags
+        self._states = None
+        self._resets = __import__('numpy').zeros(self._env_num)
         self._total_envstep_count = 0
         self._total_episode_count = 0
         self._total_train_sample_count = 0
@@ -129,6 +132,12 @@ class SampleSerialCollector(ISerialCollector):
             if random_collect:
                 actions = self._env.random_action()
                 policy_output = {i: {'action': torch.as_tensor(a)} for i, a in actions.items()}
+            elif str(self._policy_cfg.type).startswith('dreamer'):
+                # thread the RSSM filter state through the policy
+                policy_output = self._policy.forward(
+                    obs_t, **policy_kwargs, reset=self._resets, state=self._states
+                )
+                self._states = [policy_output[i]['state'] for i in sorted(policy_output)]
             else:
                 policy_output = self._policy.forward(obs_t, **policy_kwargs)
             self._obs_pool = obs_t
@@ -148,6 +157,7 @@ class SampleSerialCollector(ISerialCollector):
                 transition = self._policy.process_transition(self._obs_pool[env_id], policy_output[env_id], ts)
                 transition = EasyDict(transition)
                 transition.collect_iter = train_iter
+                self._resets[env_id] = float(bool(timestep.done))
                 self._traj_buffer[env_id].append(transition)
                 self._env_info[env_id]['step'] += 1
                 self._total_envstep_count += 1

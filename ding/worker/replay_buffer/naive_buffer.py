@@ -246,3 +246,39 @@ def create_buffer(cfg: EasyDict, tb_logger=None, exp_name: str = 'default_experi
 
 def get_buffer_cls(cfg: EasyDict) -> type:
     return BUFFER_REGISTRY.get(cfg.get('type', 'naive'))
+
+
+@BUFFER_REGISTRY.register('sequence')
+class SequenceReplayBuffer(NaiveReplayBuffer):
+    """FIFO buffer sampling CONSECUTIVE windows for sequence models
+    (DreamerV3 world-model training).
+
+    Parity: reference ding/worker/replay_buffer/naive_buffer.py
+    SequenceNaiveReplayBuffer ('sequence':471): sample(batch, sequence, iter)
+    -> list of ``batch`` lists, each ``sequence`` consecutive transitions.
+    Windows never straddle the ring's write head (stale/fresh seam).
+    """
+
+    def sample(self, batch: int, sequence: int = 1, cur_learner_iter: int = -1,
+               sample_range=None, replace: bool = False):
+        if batch == 0:
+            return []
+        with self._lock:
+            if self._valid_count < batch * sequence:
+                return None
+            n = self._valid_count
+            # valid start positions: window fully inside [0, n) and, when the
+            # ring has wrapped, not crossing the write head at self._tail
+            starts = []
+            tries = 0
+            while len(starts) < batch and tries < 100 * batch:
+                tries += 1
+                s = np.random.randint(0, n - sequence + 1)
+                if self._valid_count == self._replay_buffer_size:
+                    # ring wrapped: logical order starts at tail
+                    s = (self._tail + s) % n
+                    end = s + sequence
+                    if end > n:
+                        continue  # physical wrap inside window: skip
+                starts.append(s)
+            return [[self._data[(s + j) % n] for j in range(sequence)] for s in starts]

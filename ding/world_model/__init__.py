@@ -4,3 +4,4 @@ from .base_world_model import (
 )
 from .mbpo import MBPOWorldModel, EnsembleModel, EnsembleFC
 from .ddppo import DDPPOWorldMode
+from .dreamer import DREAMERWorldModel, RSSM, ConvDecoder

@@ -1,0 +1,365 @@
+"""DreamerV3 world model: RSSM latent dynamics + reconstruction / reward /
+discount heads, trained on [B, T] sequence batches.
+
+Parity: reference ding/world_model/dreamer.py (DREAMERWorldModel:16) and
+ding/world_model/model/networks.py (RSSM:14, ConvDecoder:298). Re-designed:
+the broken `shared`/`rec_depth` reference knobs are dropped, modules follow
+`.to(device)` instead of threading device strings, and the sequence batch is
+assembled with ONE stack per key (resident tensors, no per-step python loops
+beyond the inherent RSSM recurrence).
+"""
+import copy
+from typing import Any, Dict, Optional, Tuple
+
+import torch
+import torch.distributions as torchd
+import torch.nn as nn
+
+from ding.torch_utils.network.dreamer import (
+    ContDist, DenseHead, GRUCellLN, OneHotDist, static_scan, weight_init, uniform_weight_init,
+)
+from ding.utils import WORLD_MODEL_REGISTRY, EasyDict, deep_merge_dicts
+from ding.utils.data import default_collate
+from .base_world_model import WorldModel
+
+
+class RSSM(nn.Module):
+    """Recurrent state-space model: deterministic GRU path + stochastic
+    (discrete-categorical or gaussian) latent, posterior from embeddings."""
+
+    def __init__(
+        self,
+        stoch: int = 32,
+        deter: int = 512,
+        hidden: int = 512,
+        discrete: int = 32,
+        action_size: int = 6,
+        embed_size: int = 512,
+        action_type: str = 'discrete',
+        min_std: float = 0.1,
+        unimix_ratio: float = 0.01,
+        temp_post: bool = True,
+    ):
+        super().__init__()
+        self._stoch, self._deter, self._hidden = stoch, deter, hidden
+        self._discrete = discrete
+        self._min_std = min_std
+        self._unimix_ratio = unimix_ratio
+        self._temp_post = temp_post
+        self._action_type = action_type
+        stoch_dim = stoch * discrete if discrete else stoch
+        act = nn.SiLU
+        self._inp_layers = nn.Sequential(
+            nn.Linear(stoch_dim + action_size, hidden, bias=False), nn.LayerNorm(hidden, eps=1e-3), act()
+        )
+        self._cell = GRUCellLN(hidden, deter, norm=True)
+        self._img_out_layers = nn.Sequential(
+            nn.Linear(deter, hidden, bias=False), nn.LayerNorm(hidden, eps=1e-3), act()
+        )
+        obs_in = deter + embed_size if temp_post else embed_size
+        self._obs_out_layers = nn.Sequential(
+            nn.Linear(obs_in, hidden, bias=False), nn.LayerNorm(hidden, eps=1e-3), act()
+        )
+        stat_out = stoch * discrete if discrete else 2 * stoch
+        self._ims_stat_layer = nn.Linear(hidden, stat_out)
+        self._obs_stat_layer = nn.Linear(hidden, stat_out)
+        self.apply(weight_init)
+        uniform_weight_init(1.0)(self._ims_stat_layer)
+        uniform_weight_init(1.0)(self._obs_stat_layer)
+
+    @property
+    def feat_size(self) -> int:
+        return (self._stoch * self._discrete if self._discrete else self._stoch) + self._deter
+
+    def initial(self, batch_size: int, device=None) -> Dict[str, torch.Tensor]:
+        device = device or next(self.parameters()).device
+        deter = torch.zeros(batch_size, self._deter, device=device)
+        if self._discrete:
+            return dict(
+                logit=torch.zeros(batch_size, self._stoch, self._discrete, device=device),
+                stoch=torch.zeros(batch_size, self._stoch, self._discrete, device=device),
+                deter=deter,
+            )
+        return dict(
+            mean=torch.zeros(batch_size, self._stoch, device=device),
+            std=torch.zeros(batch_size, self._stoch, device=device),
+            stoch=torch.zeros(batch_size, self._stoch, device=device),
+            deter=deter,
+        )
+
+    def observe(self, embed: torch.Tensor, action: torch.Tensor, state: Optional[dict] = None):
+        """embed/action: [B, T, ...] -> (post, prior) with [B, T, ...] stats."""
+        swap = lambda x: x.permute([1, 0] + list(range(2, len(x.shape))))
+        if state is None:
+            state = self.initial(action.shape[0], action.device)
+        embed, action = swap(embed), swap(action)
+        post, prior = static_scan(
+            lambda prev, a, e: self.obs_step(prev[0], a, e), (action, embed), (state, state)
+        )
+        post = {k: swap(v) for k, v in post.items()}
+        prior = {k: swap(v) for k, v in prior.items()}
+        return post, prior
+
+    def imagine(self, action: torch.Tensor, state: Optional[dict] = None) -> dict:
+        swap = lambda x: x.permute([1, 0] + list(range(2, len(x.shape))))
+        if state is None:
+            state = self.initial(action.shape[0], action.device)
+        prior = static_scan(self.img_step, [swap(action)], state)[0]
+        return {k: swap(v) for k, v in prior.items()}
+
+    def get_feat(self, state: dict) -> torch.Tensor:
+        stoch = state['stoch']
+        if self._discrete:
+            stoch = stoch.reshape(list(stoch.shape[:-2]) + [self._stoch * self._discrete])
+        return torch.cat([stoch, state['deter']], -1)
+
+    def get_dist(self, state: dict):
+        if self._discrete:
+            return torchd.independent.Independent(OneHotDist(state['logit'], unimix_ratio=self._unimix_ratio), 1)
+        return ContDist(torchd.independent.Independent(torchd.normal.Normal(state['mean'], state['std']), 1))
+
+    def obs_step(self, prev_state, prev_action, embed, sample: bool = True):
+        if self._action_type == 'continuous':
+            prev_action = prev_action * (1.0 / torch.clip(torch.abs(prev_action), min=1.0)).detach()
+        prior = self.img_step(prev_state, prev_action, sample=sample)
+        x = torch.cat([prior['deter'], embed], -1) if self._temp_post else embed
+        x = self._obs_out_layers(x)
+        stats = self._suff_stats('obs', x)
+        stoch = self.get_dist(stats).sample() if sample else self.get_dist(stats).mode()
+        post = {'stoch': stoch, 'deter': prior['deter'], **stats}
+        return post, prior
+
+    def img_step(self, prev_state, prev_action, sample: bool = True):
+        if self._action_type == 'continuous':
+            prev_action = prev_action * (1.0 / torch.clip(torch.abs(prev_action), min=1.0)).detach()
+        prev_stoch = prev_state['stoch']
+        if self._discrete:
+            prev_stoch = prev_stoch.reshape(list(prev_stoch.shape[:-2]) + [self._stoch * self._discrete])
+        x = self._inp_layers(torch.cat([prev_stoch, prev_action], -1))
+        deter, _ = self._cell(x, [prev_state['deter']])
+        x = self._img_out_layers(deter)
+        stats = self._suff_stats('ims', x)
+        stoch = self.get_dist(stats).sample() if sample else self.get_dist(stats).mode()
+        return {'stoch': stoch, 'deter': deter, **stats}
+
+    def _suff_stats(self, name: str, x: torch.Tensor) -> dict:
+        x = self._ims_stat_layer(x) if name == 'ims' else self._obs_stat_layer(x)
+        if self._discrete:
+            return {'logit': x.reshape(list(x.shape[:-1]) + [self._stoch, self._discrete])}
+        mean, std = torch.split(x, [self._stoch] * 2, -1)
+        std = 2 * torch.sigmoid(std / 2) + self._min_std
+        return {'mean': mean, 'std': std}
+
+    def kl_loss(self, post, prior, forward: bool, free: float, lscale: float, rscale: float):
+        """KL-balanced loss with free bits: lscale*KL(sg(rhs)||lhs-ish)."""
+        kld = torchd.kl.kl_divergence
+        dist = lambda x: self.get_dist(x) if self._discrete else self.get_dist(x)._dist
+        sg = lambda x: {k: v.detach() for k, v in x.items()}
+        lhs, rhs = (prior, post) if forward else (post, prior)
+        value_lhs = value = kld(dist(lhs), dist(sg(rhs)))
+        value_rhs = kld(dist(sg(lhs)), dist(rhs))
+        loss_lhs = torch.mean(torch.clip(value_lhs, min=free))
+        loss_rhs = torch.mean(torch.clip(value_rhs, min=free))
+        return lscale * loss_lhs + rscale * loss_rhs, value, loss_lhs, loss_rhs
+
+
+class ConvDecoder(nn.Module):
+    """Feature -> image decoder (transposed convs, 'same'-style sizes)."""
+
+    def __init__(self, feat_size: int, depth: int, shape: Tuple[int, int, int], kernels=(3, 3, 3, 3)):
+        super().__init__()
+        self._shape = shape
+        self._depth = depth
+        layer_num = len(kernels)
+        self._embed_hw = shape[1] // (2 ** layer_num)
+        self._embed_c = depth * (2 ** (layer_num - 1))
+        self._linear = nn.Linear(feat_size, self._embed_c * self._embed_hw * self._embed_hw)
+        layers = []
+        c = self._embed_c
+        for i, k in enumerate(kernels):
+            out_c = shape[0] if i == layer_num - 1 else depth * (2 ** (layer_num - 2 - i))
+            layers.append(nn.ConvTranspose2d(c, out_c, k, stride=2, padding=k // 2, output_padding=1))
+            if i != layer_num - 1:
+                layers.append(nn.SiLU())
+            c = out_c
+        self._net = nn.Sequential(*layers)
+        self.apply(weight_init)
+
+    def __call__(self, features: torch.Tensor):
+        x = self._linear(features)
+        x = x.reshape(-1, self._embed_c, self._embed_hw, self._embed_hw)
+        x = self._net(x)
+        mean = x.reshape(list(features.shape[:-1]) + list(self._shape))
+        return ContDist(torchd.independent.Independent(torchd.normal.Normal(mean, 1.0), len(self._shape)))
+
+
+@WORLD_MODEL_REGISTRY.register('dreamer')
+class DREAMERWorldModel(WorldModel, nn.Module):
+    """RSSM + heads; `train` consumes [B][T] transition sequences from the
+    buffer, returns the posterior (detached) and rollout context for the
+    policy's imagination phase."""
+
+    config = dict(
+        train_freq=2,
+        eval_freq=int(1e9),
+        cuda=False,
+        model=dict(
+            state_size=4,
+            obs_type='vector',
+            action_size=2,
+            action_type='discrete',
+            encoder_hidden_size_list=[128, 128],
+            dyn_stoch=16,
+            dyn_deter=96,
+            dyn_hidden=96,
+            dyn_discrete=16,
+            unimix_ratio=0.01,
+            reward_layers=2,
+            discount_layers=2,
+            image_dec_layers=2,
+            units=128,
+            cnn_depth=32,
+            encoder_kernels=(4, 4, 4, 4),
+            decoder_kernels=(4, 4, 4, 4),
+            reward_head='twohot_symlog',
+            pred_discount=True,
+            grad_heads=('image', 'reward', 'discount'),
+            kl_forward=False,
+            kl_free=1.0,
+            kl_lscale=0.1,
+            kl_rscale=0.5,
+            model_lr=1e-4,
+            grad_clip=100,
+            batch_size=16,
+            batch_length=16,
+        ),
+    )
+
+    def __init__(self, cfg: EasyDict, env=None, tb_logger=None):
+        cfg = EasyDict(deep_merge_dicts(EasyDict(copy.deepcopy(self.config)), cfg or EasyDict({})))
+        WorldModel.__init__(self, cfg, env, tb_logger)
+        nn.Module.__init__(self)
+        m = cfg.model
+        self._m = m
+        self.pretrain_flag = True
+        self.action_size = m.action_size
+        self.action_type = m.action_type
+        self.obs_type = m.obs_type
+        if m.obs_type == 'vector':
+            from ding.model.common.encoder import FCEncoder
+            self.encoder = FCEncoder(m.state_size, list(m.encoder_hidden_size_list), activation='silu')
+            self.embed_size = m.encoder_hidden_size_list[-1]
+        else:
+            from ding.model.common.encoder import ConvEncoder
+            ehsl = [m.cnn_depth * (2 ** i) for i in range(len(m.encoder_kernels))]
+            self.encoder = ConvEncoder(
+                m.state_size, hidden_size_list=ehsl + [512], activation='silu',
+                kernel_size=list(m.encoder_kernels), stride=[2] * len(m.encoder_kernels),
+                padding=[1] * len(m.encoder_kernels),
+            )
+            self.embed_size = 512
+        self.dynamics = RSSM(
+            stoch=m.dyn_stoch, deter=m.dyn_deter, hidden=m.dyn_hidden, discrete=m.dyn_discrete,
+            action_size=m.action_size, embed_size=self.embed_size, action_type=m.action_type,
+            unimix_ratio=m.unimix_ratio,
+        )
+        feat_size = self.dynamics.feat_size
+        self.heads = nn.ModuleDict()
+        if m.obs_type == 'vector':
+            self.heads['image'] = DenseHead(
+                feat_size, (m.state_size, ), m.image_dec_layers, m.units, dist='mse', outscale=0.0
+            )
+        else:
+            self.heads['image'] = ConvDecoder(feat_size, m.cnn_depth, tuple(m.state_size), m.decoder_kernels)
+        self.heads['reward'] = DenseHead(
+            feat_size, (255, ), m.reward_layers, m.units, dist=m.reward_head, outscale=0.0
+        )
+        if m.pred_discount:
+            self.heads['discount'] = DenseHead(feat_size, [], m.discount_layers, m.units, dist='binary')
+        if self._cuda:
+            self.cuda()
+        self.optimizer = torch.optim.Adam(self.parameters(), lr=m.model_lr)
+
+    def step(self, obs, act, **kwargs):
+        raise NotImplementedError("dreamer trains the policy in latent imagination, not via step()")
+
+    def eval(self, env_buffer, envstep, train_iter):
+        pass
+
+    def should_pretrain(self) -> bool:
+        if self.pretrain_flag:
+            self.pretrain_flag = False
+            return True
+        return False
+
+    def _sequence_batch(self, env_buffer, batch_size: int, batch_length: int, train_iter: int) -> Dict[str, Any]:
+        """Sample [B] sequences of [T] transitions and stack to {k: [B,T,...]}."""
+        seqs = env_buffer.sample(batch_size, batch_length, train_iter)
+        out = {}
+        for key in ('obs', 'action', 'reward', 'done'):
+            rows = [
+                torch.stack([torch.as_tensor(t[key], dtype=torch.float32).reshape(-1) if torch.as_tensor(
+                    t[key]).dim() == 0 else torch.as_tensor(t[key], dtype=torch.float32) for t in seq])
+                for seq in seqs
+            ]
+            out[key] = torch.stack(rows)
+        return out
+
+    def train(self, env_buffer, envstep: int, train_iter: int, batch_size: Optional[int] = None,
+              batch_length: Optional[int] = None):
+        m = self._m
+        batch_size = batch_size or m.batch_size
+        batch_length = batch_length or m.batch_length
+        self.last_train_step = envstep
+        data = self._sequence_batch(env_buffer, batch_size, batch_length, train_iter)
+        data['discount'] = 1.0 - data['done'].float()
+        if self.action_type == 'continuous':
+            a = data['action']
+            data['action'] = a * (1.0 / torch.clip(torch.abs(a), min=1.0))
+            if data['action'].dim() == 2:
+                data['action'] = data['action'].unsqueeze(-1)
+        else:
+            data['action'] = torch.nn.functional.one_hot(
+                data['action'].squeeze(-1).long(), self.action_size
+            ).float()
+        data['image'] = data['obs']
+        if data['reward'].dim() == 2:
+            data['reward'] = data['reward'].unsqueeze(-1)
+        if data['discount'].dim() == 2:
+            data['discount'] = data['discount'].unsqueeze(-1)
+        device = next(self.parameters()).device
+        data = {k: v.to(device) for k, v in data.items()}
+
+        self.requires_grad_(True)
+        B, T = data['image'].shape[:2]
+        embed = self.encoder(data['image'].reshape(B * T, *data['image'].shape[2:])).reshape(B, T, -1)
+        post, prior = self.dynamics.observe(embed, data['action'])
+        kl_loss, kl_value, loss_lhs, loss_rhs = self.dynamics.kl_loss(
+            post, prior, m.kl_forward, m.kl_free, m.kl_lscale, m.kl_rscale
+        )
+        losses = {}
+        for name, head in self.heads.items():
+            feat = self.dynamics.get_feat(post)
+            if name not in m.grad_heads:
+                feat = feat.detach()
+            pred = head(feat)
+            losses[name] = -torch.mean(pred.log_prob(data[name]))
+        model_loss = sum(losses.values()) + kl_loss
+        self.optimizer.zero_grad()
+        model_loss.backward()
+        nn.utils.clip_grad_norm_(self.parameters(), m.grad_clip)
+        self.optimizer.step()
+        self.requires_grad_(False)
+
+        if self.tb_logger is not None:
+            for name, loss in losses.items():
+                self.tb_logger.add_scalar('world_model/' + name + '_loss', float(loss.detach()), envstep)
+            self.tb_logger.add_scalar('world_model/kl', float(kl_value.mean().detach()), envstep)
+        context = dict(
+            embed=embed,
+            feat=self.dynamics.get_feat(post),
+            kl=kl_value,
+            postent=self.dynamics.get_dist(post).entropy(),
+        )
+        post = {k: v.detach() for k, v in post.items()}
+        return post, context
