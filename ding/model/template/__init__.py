@@ -13,3 +13,4 @@ from .qac_dist import QACDIST
 from .ebm import EBM, DFO, LangevinMCMC
 from .vae import VanillaVAE
 from .language_transformer import LanguageTransformer
+from .hpt import HPT, PolicyStem

@@ -129,3 +129,14 @@ def test_create_model():
     cfg = dict(type='dqn', obs_shape=OBS, action_shape=ACT)
     m = create_model(cfg)
     assert isinstance(m, DQN)
+
+
+def test_hpt_model():
+    """HPT: perceiver policy stem + dueling head works as a DQN model."""
+    import torch
+    from ding.model.template import HPT
+    m = HPT(4, 3)
+    out = m(torch.randn(5, 4))
+    assert out['logit'].shape == (5, 3)
+    out['logit'].sum().backward()
+    assert m.policy_stem.tokens.grad is not None
