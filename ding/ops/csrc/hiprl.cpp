@@ -18,7 +18,7 @@ std::vector<torch::Tensor> ppo_fwd(
 std::vector<torch::Tensor> ppo_bwd(
     torch::Tensor logit_new, torch::Tensor action, torch::Tensor value_new, torch::Tensor value_old,
     torch::Tensor adv, torch::Tensor ret, torch::Tensor weight, torch::Tensor fwd_out, double clip_ratio,
-    double grad_policy, double grad_value, double grad_entropy
+    torch::Tensor grad_scales
 );
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {

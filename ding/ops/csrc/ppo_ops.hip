@@ -117,12 +117,16 @@ __global__ void ppo_bwd_kernel(
     int B,
     int N,
     float clip_ratio,
-    float grad_policy,                     // upstream grad of policy_loss
-    float grad_value,                      // upstream grad of value_loss
-    float grad_entropy                     // upstream grad of entropy_loss
+    const float* __restrict__ gscales      // [3] device: upstream grads of
+                                           // (policy, value, entropy) losses —
+                                           // device-resident so the launch is
+                                           // hipGraph-capture safe (no sync)
 ) {
     int b = blockIdx.x * blockDim.x + threadIdx.x;
     if (b >= B) return;
+    float grad_policy = gscales[0];
+    float grad_value = gscales[1];
+    float grad_entropy = gscales[2];
     const float* ln = logit_new + (int64_t)b * N;
     const float* o = fwd_out + (int64_t)b * 8;
     float* dl = d_logit + (int64_t)b * N;
@@ -188,7 +192,7 @@ std::vector<torch::Tensor> ppo_fwd(
 std::vector<torch::Tensor> ppo_bwd(
     torch::Tensor logit_new, torch::Tensor action, torch::Tensor value_new, torch::Tensor value_old,
     torch::Tensor adv, torch::Tensor ret, torch::Tensor weight, torch::Tensor fwd_out, double clip_ratio,
-    double grad_policy, double grad_value, double grad_entropy
+    torch::Tensor grad_scales
 ) {
     CHECK_INPUT(logit_new);
     int B = logit_new.size(0), N = logit_new.size(1);
@@ -201,7 +205,7 @@ std::vector<torch::Tensor> ppo_bwd(
         logit_new.data_ptr<float>(), action.data_ptr<int64_t>(), value_new.data_ptr<float>(),
         value_old.data_ptr<float>(), adv.data_ptr<float>(), ret.data_ptr<float>(), w_ptr,
         fwd_out.data_ptr<float>(), d_logit.data_ptr<float>(), d_value.data_ptr<float>(), B, N,
-        (float)clip_ratio, (float)grad_policy, (float)grad_value, (float)grad_entropy
+        (float)clip_ratio, grad_scales.data_ptr<float>()
     );
     HIP_CHECK_LAST();
     return {d_logit, d_value};

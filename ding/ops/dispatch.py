@@ -125,10 +125,14 @@ class _FusedPPODiscrete(torch.autograd.Function):
     def backward(ctx, g_policy, g_value, g_entropy, g_kl=None, g_clip=None):
         ext = _load()
         logit_new, action, value_new, value_old, adv, ret, w, out = ctx.saved_tensors
+        # upstream grad scales stay DEVICE-resident ([3] tensor) so the
+        # backward launch is hipGraph-capture safe and sync-free
+        gs = torch.stack([
+            g_policy.reshape(()), g_value.reshape(()), g_entropy.reshape(())
+        ]).float().contiguous()
         d_logit, d_value = ext.ppo_bwd(
             logit_new.contiguous(), action.contiguous(), value_new.contiguous(), value_old.contiguous(),
-            adv.contiguous(), ret.contiguous(), w, out, ctx.clip_ratio,
-            float(g_policy), float(g_value), float(g_entropy)
+            adv.contiguous(), ret.contiguous(), w, out, ctx.clip_ratio, gs
         )
         return d_logit, d_value, None, None, None, None, None, None, None, None
 
