@@ -14,3 +14,4 @@ from .ebm import EBM, DFO, LangevinMCMC
 from .vae import VanillaVAE
 from .language_transformer import LanguageTransformer
 from .hpt import HPT, PolicyStem
+from .diffusion import PlanDiffuser, GaussianDiffusion, ValueDiffusion, GaussianInvDynDiffusion

@@ -21,6 +21,7 @@ from .happo import HAPPOPolicy
 from . import command_mode_policy_instance  # registers '<name>_command' variants
 from .mbpolicy import MBSACPolicy, STEVESACPolicy
 from .dreamer import DREAMERPolicy
+from .plan_diffuser import PDPolicy
 # late-registered policies (dreamer/mbpolicy) also need '_command' variants
 from .command_mode_policy_instance import _register_command_variants as _rcv
 _rcv()

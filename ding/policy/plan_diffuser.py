@@ -105,11 +105,16 @@ class PDPolicy(Policy):
         conds = {}
         vals = data['condition_val']
         ids = data['condition_id']
-        for i in range(len(ids)):
-            key = ids[i][0].item() if isinstance(ids[i], torch.Tensor) else int(ids[i][0])
-            conds[key] = vals[i]
-        if len(ids) > 1:
-            self.use_target = True
+        if isinstance(ids, torch.Tensor) and ids.dim() == 1:
+            # one condition slot per sample (collated to [B]): e.g. {0: s_0}
+            conds[int(ids[0])] = vals
+        else:
+            # multiple slots: ids[i] is the i-th slot's [B] ids, vals[i] its values
+            for i in range(len(ids)):
+                key = ids[i][0].item() if isinstance(ids[i], torch.Tensor) else int(ids[i][0])
+                conds[key] = vals[i]
+            if len(ids) > 1:
+                self.use_target = True
         data['conditions'] = conds
         if 'returns' in data.keys() and data['returns'].dim() == 1:
             data['returns'] = data['returns'].unsqueeze(-1)

@@ -129,11 +129,12 @@ class DiffusionUNet1d(nn.Module):
         self.mid_block1 = ResidualTemporalBlock(mid_dim, mid_dim, embed_dim, kernel_size)
         self.mid_block2 = ResidualTemporalBlock(mid_dim, mid_dim, embed_dim, kernel_size)
         for i, (c_in, c_out) in enumerate(reversed(in_out[1:])):
-            last = i >= n - 2
+            # every downsample needs a mirroring upsample so the output
+            # horizon matches the input trajectory length
             self.ups.append(nn.ModuleList([
                 ResidualTemporalBlock(c_out * 2, c_in, embed_dim, kernel_size),
                 ResidualTemporalBlock(c_in, c_in, embed_dim, kernel_size),
-                nn.ConvTranspose1d(c_in, c_in, 4, 2, 1) if not last else nn.Identity(),
+                nn.ConvTranspose1d(c_in, c_in, 4, 2, 1),
             ]))
         self.final_conv = nn.Sequential(
             Conv1dBlock(dim, dim, kernel_size), nn.Conv1d(dim, transition_dim, 1)
