@@ -15,3 +15,4 @@ from .vae import VanillaVAE
 from .language_transformer import LanguageTransformer
 from .hpt import HPT, PolicyStem
 from .diffusion import PlanDiffuser, GaussianDiffusion, ValueDiffusion, GaussianInvDynDiffusion
+from .qgpo import QGPO

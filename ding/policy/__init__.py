@@ -22,6 +22,7 @@ from . import command_mode_policy_instance  # registers '<name>_command' variant
 from .mbpolicy import MBSACPolicy, STEVESACPolicy
 from .dreamer import DREAMERPolicy
 from .plan_diffuser import PDPolicy
+from .qgpo import QGPOPolicy
 # late-registered policies (dreamer/mbpolicy) also need '_command' variants
 from .command_mode_policy_instance import _register_command_variants as _rcv
 _rcv()

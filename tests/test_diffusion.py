@@ -111,3 +111,43 @@ def test_decision_diffuser_dd():
     # conditioned sampling with classifier-free guidance
     sample = m.conditional_sample({0: torch.randn(2, OBS)}, returns=torch.randn(2, 1))
     assert sample.trajectories.shape == (2, HORIZON, OBS)
+
+
+def test_qgpo_model_and_policy():
+    from ding.model.template.qgpo import QGPO, marginal_prob_std, dpm_solver_sample
+    from ding.policy import create_policy
+    from ding.policy.qgpo import QGPOPolicy
+    from ding.utils import deep_merge_dicts
+    obs_dim, act_dim, B, M = 4, 2, 8, 4
+    cfg = EasyDict(dict(obs_dim=obs_dim, action_dim=act_dim, qgpo_critic=dict(alpha=3, q_alpha=1)))
+    m = QGPO(cfg)
+    # marginal prob std sane: alpha->1, std->0 at t->0
+    a0, s0 = marginal_prob_std(torch.tensor([1e-4]))
+    assert float(a0) > 0.99 and float(s0) < 0.05
+    # sampling produces finite actions
+    states = np.random.randn(3, obs_dim).astype(np.float32)
+    acts = m.select_actions(states, diffusion_steps=4)
+    assert len(acts) == 3 and acts[0].shape == (act_dim, )
+    support = m.sample(states, sample_per_state=M, diffusion_steps=3)
+    assert support.shape == (3, M, act_dim)
+    # policy learn phases
+    pcfg = EasyDict(deep_merge_dicts(QGPOPolicy.default_config(), EasyDict(dict(
+        type='qgpo', cuda=False,
+        model=dict(type='qgpo', import_names=['ding.model.template.qgpo'],
+                   obs_dim=obs_dim, action_dim=act_dim, qgpo_critic=dict(alpha=3, q_alpha=1)),
+        learn=dict(learning_rate=1e-4, batch_size=B, behavior_policy_stop_training_iter=1,
+                   energy_guided_policy_begin_training_iter=1, q_value_stop_training_iter=3),
+        eval=dict(guidance_scale=[0.0, 1.0], diffusion_steps=3),
+    ))))
+    pol = create_policy(pcfg, enable_field=['learn', 'eval'])
+    data = dict(
+        s=torch.randn(B, obs_dim), a=torch.randn(B, act_dim), r=torch.randn(B, 1),
+        s_=torch.randn(B, obs_dim), d=torch.zeros(B, 1),
+        fake_a=torch.randn(B, M, act_dim), fake_a_=torch.randn(B, M, act_dim),
+    )
+    info1 = pol._forward_learn(data)  # behavior phase
+    assert info1['behavior_model_training_loss'] > 0
+    info2 = pol._forward_learn(data)  # energy phase (q0 + qt)
+    assert info2['q0_loss'] != 0 and info2['qt_loss'] != 0
+    out = pol._forward_eval({0: torch.randn(obs_dim), 1: torch.randn(obs_dim)}, guidance_scale=1.0)
+    assert out[0]['action'].shape == (act_dim, )
