@@ -16,3 +16,4 @@ from .language_transformer import LanguageTransformer
 from .hpt import HPT, PolicyStem
 from .diffusion import PlanDiffuser, GaussianDiffusion, ValueDiffusion, GaussianInvDynDiffusion
 from .qgpo import QGPO
+from .atoc import ATOC

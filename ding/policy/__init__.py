@@ -23,6 +23,7 @@ from .mbpolicy import MBSACPolicy, STEVESACPolicy
 from .dreamer import DREAMERPolicy
 from .plan_diffuser import PDPolicy
 from .qgpo import QGPOPolicy
+from .atoc import ATOCPolicy
 # late-registered policies (dreamer/mbpolicy) also need '_command' variants
 from .command_mode_policy_instance import _register_command_variants as _rcv
 _rcv()

@@ -406,3 +406,105 @@ class ProcedureCloningBFSPolicy(BehaviourCloningPolicy):
         learn=dict(batch_size=32, learning_rate=1e-3, update_per_collect=1, weight_decay=1e-4,
                    ce_label_smooth=False, show_accuracy=False, tanh_mask=False, lr_decay=False, momentum=0.9),
     )
+
+
+@POLICY_REGISTRY.register('IL')
+class ILPolicy(Policy):
+    """Imitation learning by logit regression: supervise the learner's action
+    logits against recorded expert logits (MSE).
+
+    Parity: reference ding/policy/il.py ('IL':15) — generalised: the
+    reference hard-codes a football expert; here any expert data with
+    'obs'/'logit' fields trains any logit-producing model.
+    """
+
+    config = dict(
+        type='IL',
+        cuda=False,
+        on_policy=False,
+        priority=False,
+        model=dict(),
+        learn=dict(multi_gpu=False, update_per_collect=20, batch_size=64, learning_rate=0.0002),
+        collect=dict(unroll_len=1, discount_factor=0.99),
+        eval=dict(evaluator=dict(eval_freq=800)),
+        other=dict(replay_buffer=dict(replay_buffer_size=100000)),
+    )
+
+    def default_model(self) -> tuple:
+        return 'dqn', ['ding.model.template.q_learning']
+
+    def _init_learn(self) -> None:
+        self._optimizer = Adam(self._model.parameters(), lr=self._cfg.learn.learning_rate)
+        self._learn_model = model_wrap(self._model, wrapper_name='base')
+        self._learn_model.train()
+        self._learn_model.reset()
+        self._forward_learn_cnt = 0
+
+    def _forward_learn(self, data) -> Dict[str, Any]:
+        if isinstance(data, list):
+            data = default_collate(data, cat_1dim=False)
+        if self._cuda:
+            data = to_device(data, self._device)
+        obs = data['obs']
+        if isinstance(obs, dict):
+            obs = obs.get('processed_obs', next(iter(obs.values())))
+        logit = data['logit']
+        model_logit = self._learn_model.forward(obs.float())['logit']
+        supervised_loss = torch.nn.functional.mse_loss(model_logit, logit)
+        self._optimizer.zero_grad()
+        supervised_loss.backward()
+        self._optimizer.step()
+        self._forward_learn_cnt += 1
+        return {'cur_lr': self._optimizer.defaults['lr'], 'supervised_loss': supervised_loss.item()}
+
+    def _monitor_vars_learn(self):
+        return ['cur_lr', 'supervised_loss']
+
+    def _state_dict_learn(self) -> Dict[str, Any]:
+        return {'model': self._learn_model.state_dict(), 'optimizer': self._optimizer.state_dict()}
+
+    def _load_state_dict_learn(self, state_dict: Dict[str, Any]) -> None:
+        self._learn_model.load_state_dict(state_dict['model'])
+        self._optimizer.load_state_dict(state_dict['optimizer'])
+
+    def _init_collect(self) -> None:
+        self._collect_model = model_wrap(self._model, wrapper_name='argmax_sample')
+        self._collect_model.eval()
+        self._collect_model.reset()
+        self._gamma = self._cfg.collect.discount_factor
+
+    def _forward_collect(self, data: dict, **kwargs) -> dict:
+        data_id = list(data.keys())
+        obs = default_collate(list(data.values()))
+        if self._cuda:
+            obs = to_device(obs, self._device)
+        with torch.no_grad():
+            output = self._collect_model.forward(obs.float())
+        if self._cuda:
+            output = to_device(output, 'cpu')
+        output = default_decollate(output)
+        return {i: d for i, d in zip(data_id, output)}
+
+    def _process_transition(self, obs, model_output, timestep) -> Dict[str, Any]:
+        return {
+            'obs': obs, 'next_obs': timestep.obs, 'logit': model_output['logit'],
+            'action': model_output['action'], 'reward': timestep.reward, 'done': timestep.done,
+        }
+
+    def _get_train_sample(self, data):
+        from ding.rl_utils import get_train_sample
+        return get_train_sample(data, self._cfg.collect.unroll_len)
+
+    def _init_eval(self) -> None:
+        self._eval_model = model_wrap(self._model, wrapper_name='argmax_sample')
+        self._eval_model.eval()
+        self._eval_model.reset()
+
+    def _forward_eval(self, data: dict) -> dict:
+        return self._forward_collect(data)
+
+
+# reference name alias: 'td3-vae' (dash) == 'td3_vae'
+from ding.utils import POLICY_REGISTRY as _PR
+if 'td3-vae' not in _PR:
+    _PR.register('td3-vae')(TD3VAEPolicy)

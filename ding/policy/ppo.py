@@ -553,3 +553,62 @@ class PPOOffPolicy(PPOPolicy):
             transitions, done=transitions[-1]['done'], gamma=self._gamma, gae_lambda=self._gae_lambda, cuda=False
         )
         return get_train_sample(data, self._unroll_len)
+
+
+@POLICY_REGISTRY.register('ppo_stdim')
+class PPOSTDIMPolicy(PPOPolicy):
+    """PPO + ST-DIM auxiliary contrastive representation loss over the
+    actor-critic encoder.
+
+    Parity: reference ding/policy/ppo.py PPOSTDIMPolicy ('ppo_stdim':1591).
+    """
+
+    config = dict(
+        type='ppo_stdim',
+        aux_loss_weight=0.001,
+    )
+
+    def _init_learn(self) -> None:
+        super()._init_learn()
+        from ding.torch_utils.loss import ContrastiveLoss
+        x_size, y_size = self._get_encoding_size()
+        self._aux_model = ContrastiveLoss(x_size, y_size)
+        if self._cuda:
+            self._aux_model.cuda()
+        self._aux_optimizer = Adam(self._aux_model.parameters(), lr=self._cfg.learn.learning_rate)
+        self._aux_loss_weight = self._cfg.aux_loss_weight
+
+    def _encoder(self):
+        # VAC: shared encoder or actor-side encoder
+        enc = getattr(self._model, 'encoder', None)
+        if enc is None:
+            enc = self._model.actor_encoder
+        return enc
+
+    def _get_encoding_size(self):
+        obs = self._cfg.model.obs_shape
+        test = torch.randn(1, obs) if isinstance(obs, int) else torch.randn(1, *obs)
+        if self._cuda:
+            test = test.cuda()
+        with torch.no_grad():
+            x = self._encoder()(test)
+        return x.shape[1], x.shape[1]
+
+    def _forward_learn(self, data) -> List[Dict[str, Any]]:
+        collated = default_preprocess_learn(data, ignore_done=self._cfg.learn.ignore_done, use_nstep=False)
+        if self._cuda:
+            collated = to_device(collated, self._device)
+        with torch.no_grad():
+            x = self._encoder()(collated['obs'].float())
+            y = self._encoder()(collated['next_obs'].float())
+        aux_loss = self._aux_model(x, y) * self._aux_loss_weight
+        self._aux_optimizer.zero_grad()
+        aux_loss.backward()
+        self._aux_optimizer.step()
+        out = super()._forward_learn(data)
+        for info in out if isinstance(out, list) else [out]:
+            info['aux_loss'] = aux_loss.item()
+        return out
+
+    def _monitor_vars_learn(self) -> List[str]:
+        return super()._monitor_vars_learn() + ['aux_loss']

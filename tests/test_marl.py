@@ -67,3 +67,36 @@ def test_happo():
                  actor_hidden_size_list=[32, 32], critic_hidden_size_list=[32, 32])
     main, create = _marl_cfg('happo', model, extra)
     serial_pipeline_onpolicy((main, create), seed=0, max_train_iter=2)
+
+
+def test_atoc_policy():
+    """ATOC: communicating MADDPG — collect with delta_q, learn all three losses."""
+    import torch
+    from ding.policy import create_policy
+    from ding.policy.atoc import ATOCPolicy
+    from ding.utils import EasyDict, deep_merge_dicts
+    A, OBS, ACT = 3, 6, 2
+    cfg = EasyDict(deep_merge_dicts(ATOCPolicy.default_config(), EasyDict(dict(
+        type='atoc', cuda=False,
+        model=dict(type='atoc', import_names=['ding.model.template.atoc'],
+                   obs_shape=OBS, action_shape=ACT, thought_size=8, n_agent=A,
+                   communication=True, agent_per_group=2),
+        learn=dict(batch_size=8),
+        collect=dict(n_sample=8, unroll_len=1),
+    ))))
+    pol = create_policy(cfg, enable_field=['learn', 'collect', 'eval'])
+    out = pol._forward_collect({0: torch.randn(A, OBS), 1: torch.randn(A, OBS)})
+    assert out[0]['action'].shape == (A, ACT)
+    assert 'delta_q' in out[0]
+    from ding.envs import BaseEnvTimestep
+    trans = [
+        pol._process_transition(
+            torch.randn(A, OBS), out[0],
+            BaseEnvTimestep(torch.randn(A, OBS), torch.randn(1), False, {})
+        ) for _ in range(8)
+    ]
+    samples = pol._get_train_sample(trans)
+    infos = pol._forward_learn(samples)
+    assert 'critic_loss' in infos and 'actor_loss' in infos and 'attention_loss' in infos
+    ev = pol._forward_eval({0: torch.randn(A, OBS)})
+    assert ev[0]['action'].shape == (A, ACT)
