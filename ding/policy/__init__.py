@@ -24,6 +24,7 @@ from .dreamer import DREAMERPolicy
 from .plan_diffuser import PDPolicy
 from .qgpo import QGPOPolicy
 from .atoc import ATOCPolicy
+from .ppof import PPOFPolicy
 # late-registered policies (dreamer/mbpolicy) also need '_command' variants
 from .command_mode_policy_instance import _register_command_variants as _rcv
 _rcv()

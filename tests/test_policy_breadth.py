@@ -164,3 +164,31 @@ def test_d4pg_smoke():
     main, create = pendulum_cfg('d4pg', extra_policy=dict(
         nstep=2, model=dict(action_space='regression', v_min=-100, v_max=100, n_atom=51)))
     serial_pipeline((main, create), seed=0, max_train_iter=2)
+
+
+def test_ppof_policy():
+    """PPOF simplified high-level PPO: collect -> train -> eval round trip."""
+    import torch
+    from ding.policy.ppof import PPOFPolicy
+    from ding.model.template.vac import VAC
+    cfg = PPOFPolicy.default_config()
+    cfg.cuda = False
+    cfg.batch_size = 8
+    cfg.epoch_per_collect = 2
+    cfg.n_sample = 32
+    model = VAC(obs_shape=4, action_shape=2, encoder_hidden_size_list=[16, 16])
+    pol = PPOFPolicy(cfg, model)
+    out = pol.collect(torch.randn(32, 4))
+    assert out['action'].shape == (32, )
+    data = {
+        'obs': torch.randn(32, 4), 'next_obs': torch.randn(32, 4),
+        'action': out['action'], 'logit': out['logit'],
+        'reward': torch.randn(32), 'done': torch.zeros(32),
+    }
+    infos = pol.forward(data)
+    assert len(infos) == 2 * (32 // 8)
+    assert all(abs(i['total_loss']) < 1e6 for i in infos)
+    ev = pol.eval(torch.randn(5, 4))
+    assert ev['action'].shape == (5, )
+    sd = pol.state_dict()
+    pol.load_state_dict(sd)
