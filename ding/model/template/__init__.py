@@ -17,3 +17,5 @@ from .hpt import HPT, PolicyStem
 from .diffusion import PlanDiffuser, GaussianDiffusion, ValueDiffusion, GaussianInvDynDiffusion
 from .qgpo import QGPO
 from .atoc import ATOC
+from .marl_models import MADQN, WQMix, QTran, CollaQ, HAVAC
+from .maqac import DiscreteMAQAC, ContinuousMAQAC
