@@ -19,3 +19,4 @@ from .qgpo import QGPO
 from .atoc import ATOC
 from .marl_models import MADQN, WQMix, QTran, CollaQ, HAVAC
 from .maqac import DiscreteMAQAC, ContinuousMAQAC
+from .extras import SQN, PPG, BCQ, NGU, ProcedureCloningBFS, ProcedureCloningMCTS, AutoregressiveEBM, ContinuousQVAC

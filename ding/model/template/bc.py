@@ -129,3 +129,9 @@ class EDAC(nn.Module):
         x = x.repeat(1, self.ensemble_num).unsqueeze(-1)  # [B, E*(O+A), 1]
         q = self.critic(x)['pred'].view(B, self.ensemble_num).permute(1, 0)  # [E, B]
         return {'q_value': q}
+
+
+# reference also registers DiscreteBC under 'discrete_bc'
+from ding.utils import MODEL_REGISTRY as _MR
+if 'discrete_bc' not in _MR:
+    _MR.register('discrete_bc')(DiscreteBC)
