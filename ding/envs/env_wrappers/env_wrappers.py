@@ -380,3 +380,130 @@ def update_shape(obs_shape, act_shape, rew_shape, wrapper_names):
         elif name == 'frame_stack':
             obs_shape = (4, *obs_shape)
     return obs_shape, act_shape, rew_shape
+
+
+@ENV_WRAPPER_REGISTRY.register('static_obs_norm')
+class StaticObsNormWrapper(EnvWrapper):
+    """Normalize observations with a PRECOMPUTED dataset mean/std
+    (reference env_wrappers.py StaticObsNormWrapper:790)."""
+
+    def __init__(self, env, mean: np.ndarray, std: np.ndarray):
+        super().__init__(env)
+        self.mean = np.asarray(mean, dtype=np.float32)
+        self.std = np.asarray(std, dtype=np.float32)
+        self.clip_range = (-3, 3)
+
+    def _norm(self, obs):
+        return np.clip((np.asarray(obs, dtype=np.float32) - self.mean) / (self.std + 1e-8), *self.clip_range)
+
+    def reset(self, **kwargs):
+        return self._norm(self.env.reset(**kwargs))
+
+    def step(self, action):
+        obs, reward, done, info = self.env.step(action)
+        return self._norm(obs), reward, done, info
+
+
+@ENV_WRAPPER_REGISTRY.register('ram')
+class RamWrapper(EnvWrapper):
+    """Reshape a RAM vector observation into an image-like [N, 1, 1] tensor
+    (reference RamWrapper:912)."""
+
+    def reset(self, **kwargs):
+        obs = self.env.reset(**kwargs)
+        return np.asarray(obs, dtype=np.float32).reshape(-1, 1, 1)
+
+    def step(self, action):
+        obs, reward, done, info = self.env.step(action)
+        return np.asarray(obs, dtype=np.float32).reshape(-1, 1, 1), reward, done, info
+
+
+@ENV_WRAPPER_REGISTRY.register('obs_plus_prev_act_rew')
+class ObsPlusPrevActRewWrapper(EnvWrapper):
+    """NGU input contract: obs dict {obs, prev_action, prev_reward_extrinsic}
+    (reference ObsPlusPrevActRewWrapper:1144)."""
+
+    def __init__(self, env):
+        super().__init__(env)
+        self.prev_action = -1
+        self.prev_reward_extrinsic = 0.0
+
+    def reset(self, **kwargs):
+        obs = self.env.reset(**kwargs)
+        self.prev_action = -1
+        self.prev_reward_extrinsic = 0.0
+        return {'obs': obs, 'prev_action': self.prev_action, 'prev_reward_extrinsic': self.prev_reward_extrinsic}
+
+    def step(self, action):
+        obs, reward, done, info = self.env.step(action)
+        out = {'obs': obs, 'prev_action': self.prev_action, 'prev_reward_extrinsic': self.prev_reward_extrinsic}
+        self.prev_action = action
+        self.prev_reward_extrinsic = float(np.asarray(reward).reshape(-1)[0])
+        return out, reward, done, info
+
+
+@ENV_WRAPPER_REGISTRY.register('transpose')
+class TransposeWrapper(EnvWrapper):
+    """HWC -> CHW observation transpose (reference TransposeWrapper:1213)."""
+
+    def _process_obs(self, obs):
+        return np.transpose(np.asarray(obs), (2, 0, 1))
+
+    def reset(self, **kwargs):
+        return self._process_obs(self.env.reset(**kwargs))
+
+    def step(self, action):
+        obs, reward, done, info = self.env.step(action)
+        return self._process_obs(obs), reward, done, info
+
+
+@ENV_WRAPPER_REGISTRY.register('gym_to_gymnasium')
+class GymToGymnasiumWrapper(EnvWrapper):
+    """Adapt a gymnasium-API env (reset->(obs, info), step->5-tuple) to the
+    gym 4-tuple API (reference GymToGymnasiumWrapper:1436)."""
+
+    def __init__(self, env):
+        super().__init__(env)
+        self._seed = None
+
+    def seed(self, seed: int, dynamic_seed: bool = True) -> None:
+        self._seed = seed
+
+    def reset(self, **kwargs):
+        if self._seed is not None:
+            out = self.env.reset(seed=self._seed, **kwargs)
+        else:
+            out = self.env.reset(**kwargs)
+        return out[0] if isinstance(out, tuple) else out
+
+    def step(self, action):
+        out = self.env.step(action)
+        if len(out) == 5:
+            obs, reward, terminated, truncated, info = out
+            return obs, reward, terminated or truncated, info
+        return out
+
+
+@ENV_WRAPPER_REGISTRY.register('all_in_obs')
+class AllinObsWrapper(EnvWrapper):
+    """Decision-Transformer input contract: obs dict {obs, reward}
+    (reference AllinObsWrapper:1490)."""
+
+    def reset(self, **kwargs):
+        obs = self.env.reset(**kwargs)
+        return {'obs': obs, 'reward': np.zeros(1, dtype=np.float32)}
+
+    def step(self, action):
+        obs, reward, done, info = self.env.step(action)
+        return {'obs': obs, 'reward': np.asarray(reward, dtype=np.float32).reshape(-1)}, reward, done, info
+
+
+@ENV_WRAPPER_REGISTRY.register('gym_hybrid_dict_action')
+class GymHybridDictActionWrapper(EnvWrapper):
+    """Tuple -> Dict action adaptation for gym-hybrid envs
+    (reference GymHybridDictActionWrapper:1077)."""
+
+    def step(self, action):
+        if isinstance(action, dict):
+            action = (action.get('type', action.get('action_type')), action.get('args', action.get('action_args')))
+        return self.env.step(action)

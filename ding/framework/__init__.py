@@ -10,3 +10,4 @@ def ding_init(cfg):
     framework/__init__.py:10)."""
     from ding.utils import DistributedWriter
     return DistributedWriter.get_instance(cfg.exp_name + "/log")
+from .wrapper import StepTimer

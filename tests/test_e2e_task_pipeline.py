@@ -93,3 +93,30 @@ def test_cartpole_ppo_pipeline(tmp_path):
     assert task.ctx.train_iter > 0
     collector_env.close()
     evaluator_env.close()
+
+
+def test_step_timer_wrapper():
+    from ding.framework import StepTimer
+    from ding.framework import task as _task
+
+    calls = []
+
+    def mw(ctx):
+        calls.append('fwd')
+        yield
+        calls.append('bwd')
+
+    timer = StepTimer(print_per_step=1)
+    wrapped = timer(mw)
+
+    class Ctx:
+        total_step = 0
+
+    g = wrapped(Ctx())
+    next(g)
+    try:
+        next(g)
+    except StopIteration:
+        pass
+    assert calls == ['fwd', 'bwd']
+    assert len(timer.records['mw']) == 1
