@@ -81,12 +81,16 @@ class PPOPolicy(Policy):
                         torch.nn.init.zeros_(m.bias)
                         m.weight.data.copy_(0.01 * m.weight.data)
 
+        _capturable = bool(self._cfg.learn.get('cuda_graph', False)) and not self._cfg.multi_gpu \
+            and self._cuda
         self._optimizer = Adam(
             self._model.parameters(),
             lr=self._cfg.learn.learning_rate,
             grad_clip_type=self._cfg.learn.grad_clip_type,
             clip_value=self._cfg.learn.grad_clip_value,
+            capturable=_capturable,
         )
+        self._optimizer_capturable = _capturable
         self._learn_model = model_wrap(self._model, wrapper_name='base')
         self._value_weight = self._cfg.learn.value_weight
         self._entropy_weight = self._cfg.learn.entropy_weight
@@ -128,6 +132,10 @@ class PPOPolicy(Policy):
                 total_loss = policy_loss + wv * value_loss - we * entropy_loss
                 self._optimizer.zero_grad(set_to_none=False)
                 total_loss.backward()
+                if self._optimizer_capturable:
+                    # capturable Adam (on-device step counters): grad clip +
+                    # update replay inside the graph
+                    self._optimizer.step()
                 return {
                     'total_loss': total_loss.detach(), 'policy_loss': policy_loss.detach(),
                     'value_loss': value_loss.detach(), 'entropy_loss': entropy_loss.detach(),
@@ -138,7 +146,8 @@ class PPOPolicy(Policy):
 
             self._graphed_step = GraphedStep(step_fn)
         out = self._graphed_step(batch)
-        self._optimizer.step()
+        if not self._optimizer_capturable:
+            self._optimizer.step()
         return out
 
     def _forward_learn(self, data: List[Dict[str, Any]]) -> List[Dict[str, Any]]:
