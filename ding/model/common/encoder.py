@@ -48,6 +48,27 @@ class UnfoldConv2d(nn.Module):
         return out.reshape(B, self.conv.out_channels, Ho, Wo)
 
 
+class StemConv2d(nn.Conv2d):
+    """Drop-in nn.Conv2d for the Atari stem (8x8, stride 4, pad 0) that
+    routes GPU fp32 forwards through the direct HIP kernel
+    (ding/ops/csrc/conv_ops.hip) — rocprof showed MIOpen's tuned choice for
+    this shape is a per-sample im2col loop (92k Im2d2Col launches per PPO
+    bench run). State dict and numerics match nn.Conv2d (fp32 FMA order
+    differs within tolerance). Input gradients are not produced: the stem
+    is the input layer."""
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if (
+            x.is_cuda and x.dtype == torch.float32 and not x.requires_grad
+            and self.kernel_size == (8, 8) and self.stride == (4, 4) and self.padding == (0, 0)
+            and (x.shape[-1] - 8) // 4 + 1 <= 32
+        ):
+            from ding.ops import dispatch as _dispatch
+            if _dispatch.use_hip_autograd(x):
+                return _dispatch.stem_conv2d(x, self.weight, self.bias)
+        return super().forward(x)
+
+
 class ConvEncoder(nn.Module):
     """Nature-DQN style conv stack + flatten + fc to hidden_size_list[-1]."""
 
@@ -76,6 +97,10 @@ class ConvEncoder(nn.Module):
             if fast_im2col and k >= 5:
                 # large-kernel strided conv: batched unfold+GEMM (see UnfoldConv2d)
                 layers.append(UnfoldConv2d(in_c, hidden_size_list[i], k, s, p))
+                layers.append(build_activation(activation))
+            elif i == 0 and k == 8 and s == 4 and p == 0 and norm_type is None:
+                # Atari stem: direct HIP conv on GPU (see StemConv2d)
+                layers.append(StemConv2d(in_c, hidden_size_list[i], k, s, p))
                 layers.append(build_activation(activation))
             else:
                 layers.append(

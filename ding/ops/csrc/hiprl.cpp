@@ -15,6 +15,13 @@ std::vector<torch::Tensor> ppo_fwd(
     torch::Tensor value_old, torch::Tensor adv, torch::Tensor ret, torch::Tensor weight, double clip_ratio,
     int64_t use_value_clip
 );
+torch::Tensor stem_conv_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias);
+torch::Tensor stem_conv_wrw(torch::Tensor x, torch::Tensor dy, int64_t O);
+std::vector<torch::Tensor> q_nstep_fwd(
+    torch::Tensor q, torch::Tensor next_n_q, torch::Tensor action, torch::Tensor next_action,
+    torch::Tensor reward, torch::Tensor done, torch::Tensor value_gamma, double gamma, int64_t nstep,
+    int64_t rescale
+);
 std::vector<torch::Tensor> ppo_bwd(
     torch::Tensor logit_new, torch::Tensor action, torch::Tensor value_new, torch::Tensor value_old,
     torch::Tensor adv, torch::Tensor ret, torch::Tensor weight, torch::Tensor fwd_out, double clip_ratio,
@@ -29,4 +36,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("scatter_connection", &scatter_connection, "entity->spatial scatter");
     m.def("ppo_fwd", &ppo_fwd, "fused PPO loss forward");
     m.def("ppo_bwd", &ppo_bwd, "fused PPO loss backward");
+    m.def("q_nstep_fwd", &q_nstep_fwd, "fused n-step Q TD forward (+value rescale)");
+    m.def("stem_conv_fwd", &stem_conv_fwd, "direct 8x8s4 stem conv forward");
+    m.def("stem_conv_wrw", &stem_conv_wrw, "direct 8x8s4 stem conv weight grad");
 }

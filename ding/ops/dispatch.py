@@ -164,3 +164,69 @@ def c51_project(next_n_dist, next_n_act, reward_n, done, v_min, v_max, gamma_n):
         next_n_dist.contiguous().float(), next_n_act.contiguous().long(), reward_n.contiguous().float(),
         done.contiguous().float(), float(v_min), float(v_max), float(gamma_n)
     )
+
+
+class _FusedQNstepTD(torch.autograd.Function):
+    """Fused n-step Q TD: HIP forward computes {td, return, q_sa} in one
+    launch; backward is the single scatter d(td)/d(q[b,a]) = 1 (target
+    detached), expressed in eager torch."""
+
+    @staticmethod
+    def forward(ctx, q, next_n_q, action, next_action, reward, done, value_gamma, gamma, nstep, rescale):
+        ext = _load()
+        vg = value_gamma if value_gamma is not None else torch.empty(0, device=q.device)
+        (out, ) = ext.q_nstep_fwd(
+            q.detach().contiguous(), next_n_q.detach().contiguous(), action.contiguous(),
+            next_action.contiguous(), reward.detach().contiguous(), done.detach().float().contiguous(), vg,
+            float(gamma), int(nstep), int(rescale)
+        )
+        td, ret, q_sa = out[:, 0], out[:, 1], out[:, 2]
+        ctx.save_for_backward(action)
+        ctx.q_shape = q.shape
+        ctx.mark_non_differentiable(ret)
+        return td, ret
+
+    @staticmethod
+    def backward(ctx, g_td, g_ret=None):
+        (action, ) = ctx.saved_tensors
+        d_q = torch.zeros(ctx.q_shape, device=g_td.device, dtype=g_td.dtype)
+        d_q.scatter_(1, action.unsqueeze(1), g_td.unsqueeze(1))
+        return d_q, None, None, None, None, None, None, None, None, None
+
+
+def fused_q_nstep_td(q, next_n_q, action, next_action, reward, done, value_gamma, gamma, nstep,
+                     rescale: bool = False):
+    """Returns (td = q_sa - return  [differentiable wrt q], return)."""
+    return _FusedQNstepTD.apply(q, next_n_q, action, next_action, reward, done, value_gamma, gamma, nstep,
+                                1 if rescale else 0)
+
+
+class _StemConvFn(torch.autograd.Function):
+    """Direct 8x8s4 stem conv: HIP forward + HIP weight-grad; no input grad
+    (input layer). Bias grad is a cheap eager reduction."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias):
+        ext = _load()
+        y = ext.stem_conv_fwd(
+            x.detach().contiguous(), weight.detach().contiguous(),
+            bias.detach().contiguous() if bias is not None else torch.empty(0, device=x.device)
+        )
+        ctx.save_for_backward(x)
+        ctx.O = weight.shape[0]
+        ctx.has_bias = bias is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = _load()
+        (x, ) = ctx.saved_tensors
+        dy = dy.contiguous()
+        dw = ext.stem_conv_wrw(x.contiguous(), dy, ctx.O)
+        db = dy.sum(dim=(0, 2, 3)) if ctx.has_bias else None
+        return None, dw, db
+
+
+def stem_conv2d(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor = None) -> torch.Tensor:
+    """conv2d(x, w, b, stride=4) for 8x8 kernels, pad 0, W_out<=32."""
+    return _StemConvFn.apply(x, weight, bias)

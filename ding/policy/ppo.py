@@ -81,8 +81,10 @@ class PPOPolicy(Policy):
                         torch.nn.init.zeros_(m.bias)
                         m.weight.data.copy_(0.01 * m.weight.data)
 
-        _capturable = bool(self._cfg.learn.get('cuda_graph', False)) and not self._cfg.multi_gpu \
-            and self._cuda
+        # NOTE: capturable Adam inside the graph was A/B'd on MI355X: no
+        # throughput gain (219 vs 214 ms/step) and it broke graph-vs-eager
+        # equivalence; the optimizer step stays eager after replay.
+        _capturable = False
         self._optimizer = Adam(
             self._model.parameters(),
             lr=self._cfg.learn.learning_rate,
@@ -169,7 +171,11 @@ class PPOPolicy(Policy):
                         # minibatch-family shapes; huge batches fall to naive
                         # fp64-accum kernels (seen in rocprof on MI355X)
                         both = torch.cat([data['obs'], data['next_obs']], dim=0)
-                        chunks = torch.split(both, 3200, dim=0)
+                        # chunk at the learn minibatch size: those conv shapes
+                        # are already tuned (rocprof: batch-3200 critic passes
+                        # fell to naive fp64-accum convs at ~17 ms each)
+                        _chunk = max(1, int(self._cfg.learn.batch_size))
+                        chunks = torch.split(both, _chunk, dim=0)
                         values = torch.cat(
                             [self._learn_model.forward(c, mode='compute_critic')['value'] for c in chunks], dim=0
                         )

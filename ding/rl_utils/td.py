@@ -243,6 +243,19 @@ def q_nstep_td_error(
     if weight is None:
         weight = torch.ones_like(reward[0] if reward.dim() > 1 else reward)
     marl = action.dim() > 1
+    # fused HIP lane: single-launch forward {td, return, q_sa}; one-scatter
+    # backward (see ding/ops/csrc/td_ops.hip)
+    from ding.ops import dispatch as _dispatch
+    if (
+        not marl and not cum_reward and isinstance(criterion, nn.MSELoss) and isinstance(gamma, float)
+        and isinstance(q, torch.Tensor) and q.dim() == 2 and q.dtype == torch.float32
+        and reward.dim() == 2 and _dispatch.use_hip_autograd(q)
+    ):
+        td, _ret = _dispatch.fused_q_nstep_td(
+            q, next_n_q, action, next_n_action, reward, done, value_gamma, gamma, nstep, rescale=False
+        )
+        td_error_per_sample = td.pow(2)
+        return (td_error_per_sample * weight).mean(), td_error_per_sample
     if not marl:
         action_ = action.unsqueeze(-1)
     else:
@@ -279,6 +292,17 @@ def q_nstep_td_error_with_rescale(
     q, next_n_q, action, next_n_action, reward, done, weight = data
     if weight is None:
         weight = torch.ones_like(action, dtype=q.dtype)
+    from ding.ops import dispatch as _dispatch
+    if (
+        isinstance(criterion, nn.MSELoss) and isinstance(gamma, float) and trans_fn is value_transform
+        and inv_trans_fn is value_inv_transform and isinstance(q, torch.Tensor) and q.dim() == 2
+        and q.dtype == torch.float32 and reward.dim() == 2 and _dispatch.use_hip_autograd(q)
+    ):
+        td, _ret = _dispatch.fused_q_nstep_td(
+            q, next_n_q, action, next_n_action, reward, done, value_gamma, gamma, nstep, rescale=True
+        )
+        td_error_per_sample = td.pow(2)
+        return (td_error_per_sample * weight).mean(), td_error_per_sample
     q_s_a = q.gather(-1, action.unsqueeze(-1)).squeeze(-1)
     target_q_s_a = next_n_q.gather(-1, next_n_action.unsqueeze(-1)).squeeze(-1)
     target_q_s_a = inv_trans_fn(target_q_s_a)
