@@ -6,3 +6,4 @@ from .shm_buffer import ShmBuffer, ShmBufferContainer
 from .storage import Storage, FileStorage
 from .storage_loader import StorageLoader, FileStorageLoader
 from .model_loader import ModelLoader, FileModelLoader
+from .level_replay import LevelSampler

@@ -12,5 +12,5 @@ from .serial_entry_variants2 import (
     serial_pipeline_ngu, serial_pipeline_r2d3, serial_pipeline_preference_based_irl,
     serial_pipeline_preference_based_irl_onpolicy, serial_pipeline_trex, serial_pipeline_trex_onpolicy,
     serial_pipeline_guided_cost, serial_pipeline_td3_vae, serial_pipeline_onpolicy_ppg,
-    serial_pipeline_bco, serial_pipeline_pc, trex_collecting_data, serial_pipeline_dreamer,
+    serial_pipeline_bco, serial_pipeline_pc, trex_collecting_data, serial_pipeline_dreamer, serial_pipeline_plr, generate_seeds,
 )

@@ -581,3 +581,58 @@ def serial_pipeline_dreamer(
             break
     learner.call_hook('after_run')
     return policy
+
+
+def generate_seeds(num_seeds: int = 500, base: int = 0) -> list:
+    """Training level seed universe for PLR."""
+    return list(range(base, base + num_seeds))
+
+
+def serial_pipeline_plr(
+    input_cfg: Union[str, Tuple[dict, dict]],
+    seed: int = 0,
+    env_setting: Optional[List[Any]] = None,
+    model: Optional[torch.nn.Module] = None,
+    max_train_iter: int = int(1e10),
+    max_env_step: int = int(1e10),
+) -> 'Policy':  # noqa
+    """Prioritized Level Replay: on-policy training where each collect cycle
+    reseeds collector envs from the PLR level sampler (reference
+    serial_entry_plr.py)."""
+    from ding.data import LevelSampler
+    from ding.policy.common_utils import default_preprocess_learn
+    cfg, policy, collector_env, evaluator_env = _prepare(input_cfg, seed, env_setting, model)
+    learner, collector, evaluator, replay_buffer, commander = _build_workers(
+        cfg, policy, collector_env, evaluator_env
+    )
+    env_num = collector_env.env_num
+    train_seeds = generate_seeds(cfg.get('level_replay', {}).get('num_seeds', 128))
+    level_sampler = LevelSampler(
+        train_seeds, cfg.policy.model.obs_shape, cfg.policy.model.action_shape, env_num,
+        cfg.get('level_replay', EasyDict({}))
+    )
+    learner.call_hook('before_run')
+    seeds = [int(level_sampler.sample('sequential')) for _ in range(env_num)]
+    level_seeds = torch.Tensor(seeds)
+    collector_env.seed(seeds)
+    collector_env.reset()
+    while True:
+        collect_kwargs = commander.step()
+        if evaluator.should_eval(learner.train_iter):
+            stop, _ = evaluator.eval(learner.save_checkpoint, learner.train_iter, collector.envstep)
+            if stop:
+                break
+        new_data = collector.collect(
+            train_iter=learner.train_iter, level_seeds=level_seeds, policy_kwargs=collect_kwargs
+        )
+        learner.train(new_data, collector.envstep)
+        stacked = default_preprocess_learn(new_data, ignore_done=cfg.policy.learn.ignore_done, use_nstep=False)
+        level_sampler.update_with_rollouts(stacked, env_num)
+        seeds = [int(level_sampler.sample()) for _ in range(env_num)]
+        level_seeds = torch.Tensor(seeds)
+        collector_env.seed(seeds)
+        collector_env.reset()
+        if collector.envstep >= max_env_step or learner.train_iter >= max_train_iter:
+            break
+    learner.call_hook('after_run')
+    return policy

@@ -157,6 +157,9 @@ class SampleSerialCollector(ISerialCollector):
                 transition = self._policy.process_transition(self._obs_pool[env_id], policy_output[env_id], ts)
                 transition = EasyDict(transition)
                 transition.collect_iter = train_iter
+                if level_seeds is not None:
+                    # PLR: tag each transition with its env's level seed
+                    transition.seed = level_seeds[env_id]
                 self._resets[env_id] = float(bool(timestep.done))
                 self._traj_buffer[env_id].append(transition)
                 self._env_info[env_id]['step'] += 1

@@ -167,3 +167,14 @@ def test_bco_pipeline():
     main, create = cartpole_cfg('bc')
     em, ec = cartpole_cfg('dqn')
     serial_pipeline_bco((main, create), (em, ec), seed=0, max_train_iter=2)
+
+
+def test_plr_pipeline():
+    from ding.entry import serial_pipeline_plr
+    main, create = cartpole_cfg('ppo', extra_policy=dict(
+        action_space='discrete', recompute_adv=True,
+        learn=dict(epoch_per_collect=1),
+        collect=dict(discount_factor=0.99, gae_lambda=0.95),
+    ))
+    main.level_replay = EasyDict(dict(strategy='policy_entropy', num_seeds=8))
+    serial_pipeline_plr((main, create), seed=0, max_train_iter=2)
