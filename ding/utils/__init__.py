@@ -6,7 +6,7 @@ from .registry import (
     COMM_LEARNER_REGISTRY, COMM_COLLECTOR_REGISTRY, COMMANDER_REGISTRY, LEAGUE_REGISTRY, PLAYER_REGISTRY,
     MQ_REGISTRY, AGENT_REGISTRY, HOOK_REGISTRY,
 )
-from .default_helper import (
+from .default_helper import (get_task_uid, 
     set_pkg_seed, lists_to_dicts, dicts_to_lists, deep_merge_dicts, deep_update, squeeze,
     one_time_warning, error_wrapper, LimitedSpaceContainer, get_shape0, split_data_generator,
     flatten_dict, RunningMeanStd,

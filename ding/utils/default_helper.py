@@ -219,3 +219,9 @@ class RunningMeanStd:
     @property
     def std(self):
         return np.sqrt(self.var + 1e-8)
+
+
+def get_task_uid() -> str:
+    """Short unique id for parallel-pipeline task names."""
+    import uuid
+    return uuid.uuid4().hex[:8]

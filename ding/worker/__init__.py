@@ -7,3 +7,4 @@ from .coordinator.coordinator import Coordinator, ResourceManager
 from .adapter.learner_aggregator import LearnerAggregator
 from .collector.battle_collector import BattleSampleSerialCollector, BattleEpisodeSerialCollector
 from .collector.metric_serial_evaluator import MetricSerialEvaluator, IMetric
+from .coordinator.parallel_commander import BaseCommander, NaiveCommander, SoloCommander, OneVsOneCommander, create_parallel_commander
