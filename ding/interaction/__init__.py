@@ -1,1 +1,1 @@
-from .master import Master, Slave, HttpEngine, success_response, failure_response
+from .master import Master, Slave, HttpEngine, TaskFail, success_response, failure_response

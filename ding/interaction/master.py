@@ -39,6 +39,15 @@ class HttpEngine:
         return resp.json()
 
 
+class TaskFail:
+    """Marker result: the slave declines/fails a task (reference
+    ding/interaction/slave/slave.py TaskFail)."""
+
+    def __init__(self, result=None, message: str = ''):
+        self.result = result or {}
+        self.message = message
+
+
 class Slave:
     """Worker endpoint: receives tasks over POST /task/new, heartbeats to its
     master. Subclass and override ``_process_task``."""

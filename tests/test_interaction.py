@@ -52,3 +52,64 @@ def test_learner_aggregator():
     agg2 = LearnerAggregator([lambda: {'loss': 1.0}, dead])
     info2 = agg2.merge_info()
     assert info2['dead_learner_num'] == 1
+
+
+def test_comm_learner_fs_roundtrip(tmp_path):
+    """FlaskFileSystemLearner: start -> learn from FS data -> policy saved."""
+    import torch
+    from ding.worker.comm import FlaskFileSystemLearner
+    from ding.utils import EasyDict
+    from tests.test_policy_breadth import cartpole_cfg
+    from ding.config import compile_config
+
+    main, create = cartpole_cfg('dqn', extra_policy=dict(nstep=1))
+    create.policy.type = 'dqn_command'
+    cfg = compile_config(main, create_cfg=create, auto=True, save_cfg=False)
+
+    learner_worker = FlaskFileSystemLearner(
+        EasyDict({'path_data': str(tmp_path / 'data'), 'path_policy': str(tmp_path / 'policy')})
+    )
+    out = learner_worker._process_task({'name': 'resource'})
+    assert 'cpu' in out
+    out = learner_worker._process_task({
+        'name': 'learner_start_task',
+        'task_info': {'policy': dict(cfg.policy, type='dqn'), 'learner_cfg': {}, 'policy_id': 'p.pth'},
+    })
+    assert 'started' in out['message']
+    demand = learner_worker._process_task({'name': 'learner_get_data_task', 'task_id': 't0', 'buffer_id': 'b0'})
+    assert demand['batch_size'] == cfg.policy.learn.batch_size
+    # fabricate transitions on the FS
+    data = [
+        {
+            'obs': torch.randn(4), 'next_obs': torch.randn(4), 'action': torch.tensor([0]),
+            'reward': torch.randn(1), 'done': False, 'collect_iter': 0,
+        } for _ in range(demand['batch_size'])
+    ]
+    torch.save(data, tmp_path / 'data' / 'batch0.pth')
+    out = learner_worker._process_task({'name': 'learner_learn_task', 'data': ['batch0.pth']})
+    assert out['train_iter'] >= 1
+    assert (tmp_path / 'policy' / 'p.pth').exists()
+    learner_worker._process_task({'name': 'learner_close_task'})
+
+
+def test_comm_collector_fs_roundtrip(tmp_path):
+    import torch
+    from ding.worker.comm import FlaskFileSystemCollector
+    from ding.utils import EasyDict
+    from tests.test_policy_breadth import cartpole_cfg
+    from ding.config import compile_config
+
+    main, create = cartpole_cfg('dqn')
+    create.policy.type = 'dqn_command'
+    cfg = compile_config(main, create_cfg=create, auto=True, save_cfg=False)
+    cfg.policy.type = 'dqn'
+
+    worker = FlaskFileSystemCollector(EasyDict({'path_data': str(tmp_path), 'path_policy': str(tmp_path)}))
+    out = worker._process_task({'name': 'collector_start_task', 'task_info': {'cfg': cfg}})
+    assert 'started' in out['message']
+    out = worker._process_task({'name': 'collector_data_task', 'n_sample': 8,
+                                'policy_kwargs': {'eps': 0.5}})
+    assert out['sample_count'] == 8
+    data = torch.load(tmp_path / out['data_path'], weights_only=False)
+    assert len(data) == 8 and 'obs' in data[0]
+    worker._process_task({'name': 'collector_close_task'})
