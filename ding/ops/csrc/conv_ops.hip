@@ -53,6 +53,9 @@ __global__ void stem_conv_fwd_kernel(
     int ph = ph0 + prow;
     if (ph >= HO || prow >= PH_TILE) return;
 
+    // acc[] must be indexed by COMPILE-TIME constants only: a runtime-bounded
+    // `for (pw < WO)` loop demotes the accumulator array to scratch memory
+    // (measured 40x slowdown) — fully unroll with guards instead.
     float acc[MAX_WOUT];
     #pragma unroll
     for (int i = 0; i < MAX_WOUT; ++i) acc[i] = 0.f;
@@ -67,15 +70,19 @@ __global__ void stem_conv_fwd_kernel(
             #pragma unroll
             for (int kw = 0; kw < STEM_K; ++kw) {
                 float wv = wrow[kw];
-                for (int pw = 0; pw < WO; ++pw) {
-                    acc[pw] = fmaf(wv, xrow[pw * STEM_S + kw], acc[pw]);
+                #pragma unroll
+                for (int pw = 0; pw < MAX_WOUT; ++pw) {
+                    if (pw < WO) acc[pw] = fmaf(wv, xrow[pw * STEM_S + kw], acc[pw]);
                 }
             }
         }
     }
     float bv = bias ? bias[oc] : 0.f;
     float* yrow = y + (((int64_t)b * O + oc) * HO + ph) * WO;
-    for (int pw = 0; pw < WO; ++pw) yrow[pw] = acc[pw] + bv;
+    #pragma unroll
+    for (int pw = 0; pw < MAX_WOUT; ++pw) {
+        if (pw < WO) yrow[pw] = acc[pw] + bv;
+    }
 }
 
 // dW[o,c,kh,kw] = sum_{b,ph,pw} dY[b,o,ph,pw] * X[b,c,ph*4+kh,pw*4+kw]

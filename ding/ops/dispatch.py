@@ -219,10 +219,16 @@ class _StemConvFn(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dy):
-        ext = _load()
         (x, ) = ctx.saved_tensors
         dy = dy.contiguous()
-        dw = ext.stem_conv_wrw(x.contiguous(), dy, ctx.O)
+        # dW as ONE hipBLASLt GEMM: dW[O, C*64] = dY'[O, B*P] @ cols[B*P, C*64]
+        # (the HIP wrw kernel's per-entry gather measured 3.7 ms/call — the
+        # uncoalesced dW-major loop loses to im2col+GEMM here)
+        B, O = dy.shape[0], dy.shape[1]
+        cols = torch.nn.functional.unfold(x, 8, stride=4)          # [B, C*64, P]
+        dyf = dy.reshape(B, O, -1)                                 # [B, O, P]
+        dw = torch.bmm(dyf, cols.transpose(1, 2)).sum(0)           # [O, C*64]
+        dw = dw.reshape(O, x.shape[1], 8, 8)
         db = dy.sum(dim=(0, 2, 3)) if ctx.has_bias else None
         return None, dw, db
 
