@@ -15,6 +15,14 @@ std::vector<torch::Tensor> ppo_fwd(
     torch::Tensor value_old, torch::Tensor adv, torch::Tensor ret, torch::Tensor weight, double clip_ratio,
     int64_t use_value_clip
 );
+std::vector<torch::Tensor> lstm_cell_fwd(
+    torch::Tensor gxn, torch::Tensor gh_raw, torch::Tensor gamma, torch::Tensor beta, torch::Tensor bias,
+    torch::Tensor c_in
+);
+std::vector<torch::Tensor> lstm_cell_bwd(
+    torch::Tensor dh, torch::Tensor dc_next, torch::Tensor acts, torch::Tensor xhat, torch::Tensor gamma,
+    torch::Tensor c_in, torch::Tensor c_out, torch::Tensor rstd
+);
 torch::Tensor stem_conv_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias);
 torch::Tensor stem_conv_wrw(torch::Tensor x, torch::Tensor dy, int64_t O);
 std::vector<torch::Tensor> q_nstep_fwd(
@@ -39,4 +47,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("q_nstep_fwd", &q_nstep_fwd, "fused n-step Q TD forward (+value rescale)");
     m.def("stem_conv_fwd", &stem_conv_fwd, "direct 8x8s4 stem conv forward");
     m.def("stem_conv_wrw", &stem_conv_wrw, "direct 8x8s4 stem conv weight grad");
+    m.def("lstm_cell_fwd", &lstm_cell_fwd, "fused LN-LSTM cell forward");
+    m.def("lstm_cell_bwd", &lstm_cell_bwd, "fused LN-LSTM cell backward");
 }

@@ -236,3 +236,38 @@ class _StemConvFn(torch.autograd.Function):
 def stem_conv2d(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor = None) -> torch.Tensor:
     """conv2d(x, w, b, stride=4) for 8x8 kernels, pad 0, W_out<=32."""
     return _StemConvFn.apply(x, weight, bias)
+
+
+class _FusedLSTMCell(torch.autograd.Function):
+    """One-launch LN-LSTM cell: LN(gh_raw) + gates + state update; backward
+    is one kernel + eager per-parameter reductions. The gradient w.r.t.
+    gxn (input-side gates) flows back into the eager sequence-level LN_x."""
+
+    @staticmethod
+    def forward(ctx, gxn, gh_raw, gamma, beta, bias, c_in):
+        ext = _load()
+        h_out, c_out, xhat, acts, rstd = ext.lstm_cell_fwd(
+            gxn.detach().contiguous(), gh_raw.detach().contiguous(), gamma.detach().contiguous(),
+            beta.detach().contiguous(), bias.detach().contiguous(), c_in.detach().contiguous()
+        )
+        ctx.save_for_backward(acts, xhat, gamma, c_in, c_out, rstd)
+        return h_out, c_out
+
+    @staticmethod
+    def backward(ctx, dh, dc_next):
+        ext = _load()
+        acts, xhat, gamma, c_in, c_out, rstd = ctx.saved_tensors
+        dg, dgh, dc_in = ext.lstm_cell_bwd(
+            dh.contiguous(), dc_next.contiguous() if dc_next is not None else torch.empty(0, device=dh.device),
+            acts, xhat, gamma, c_in.contiguous(), c_out, rstd
+        )
+        # per-parameter reductions over the batch (single eager sums)
+        dgamma = (dg * xhat).sum(0)
+        dbeta = dg.sum(0)
+        dbias = dg.sum(0)
+        return dg, dgh, dgamma, dbeta, dbias, dc_in
+
+
+def fused_lstm_cell(gxn, gh_raw, gamma, beta, bias, c_in):
+    """Returns (h', c')."""
+    return _FusedLSTMCell.apply(gxn, gh_raw, gamma, beta, bias, c_in)

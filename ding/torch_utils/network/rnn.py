@@ -131,8 +131,21 @@ class LSTM(nn.Module, LSTMForwardWrapper):
             if self.norm_x is not None:
                 gx = self.norm_x[l](gx)
             outputs = []
+            # fused HIP lane: LN(gh) + gates + state update in ONE launch per
+            # step (vs ~13 eager kernels); see ding/ops/csrc/lstm_ops.hip
+            from ding.ops import dispatch as _dispatch
+            _fused = (
+                self.norm_h is not None and isinstance(x, torch.Tensor) and x.is_cuda
+                and x.dtype == torch.float32 and _dispatch.use_hip_autograd(x)
+            )
             for t in range(seq_len):
                 gh = torch.matmul(h, self.wh[l])
+                if _fused:
+                    h, c = _dispatch.fused_lstm_cell(
+                        gx[t], gh, self.norm_h[l].weight, self.norm_h[l].bias, self.bias[l], c
+                    )
+                    outputs.append(h)
+                    continue
                 if self.norm_h is not None:
                     gh = self.norm_h[l](gh)
                 gates = gx[t] + gh + self.bias[l]

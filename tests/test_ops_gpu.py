@@ -329,3 +329,32 @@ def test_stem_conv_in_encoder(ext):
         del os.environ['DI_ENGINE_DISABLE_HIP']
         importlib.reload(disp)
     assert torch.allclose(out_hip, out_ref, rtol=1e-3, atol=1e-3), (out_hip - out_ref).abs().max()
+
+
+def test_fused_lstm_cell_vs_eager(ext):
+    """Fused LN-LSTM matches the eager loop: outputs, states and all grads."""
+    import os, importlib
+    from ding.torch_utils.network.rnn import LSTM
+    torch.manual_seed(0)
+    T, B, I, H = 6, 5, 12, 16
+    lstm = LSTM(I, H, num_layers=2, norm_type='LN').cuda()
+    lstm2 = LSTM(I, H, num_layers=2, norm_type='LN').cuda()
+    lstm2.load_state_dict(lstm.state_dict())
+    x = torch.randn(T, B, I, device='cuda', requires_grad=True)
+    x2 = x.detach().clone().requires_grad_(True)
+    out_hip, (h_hip, c_hip) = lstm(x, None, list_next_state=False)
+    import ding.ops.dispatch as disp
+    os.environ['DI_ENGINE_DISABLE_HIP'] = '1'
+    importlib.reload(disp)
+    try:
+        out_ref, (h_ref, c_ref) = lstm2(x2, None, list_next_state=False)
+    finally:
+        del os.environ['DI_ENGINE_DISABLE_HIP']
+        importlib.reload(disp)
+    assert torch.allclose(out_hip, out_ref, rtol=1e-4, atol=1e-4), (out_hip - out_ref).abs().max()
+    assert torch.allclose(c_hip, c_ref, rtol=1e-4, atol=1e-4)
+    out_hip.sum().backward()
+    out_ref.sum().backward()
+    assert torch.allclose(x.grad, x2.grad, rtol=1e-3, atol=1e-3), (x.grad - x2.grad).abs().max()
+    for (n1, p1), (n2, p2) in zip(lstm.named_parameters(), lstm2.named_parameters()):
+        assert torch.allclose(p1.grad, p2.grad, rtol=1e-3, atol=1e-3), (n1, (p1.grad - p2.grad).abs().max())
