@@ -63,7 +63,7 @@ class StemConv2d(nn.Conv2d):
             os.environ.get('DING_STEM_CONV', '1') not in ('0', 'false')
             and x.is_cuda and x.dtype == torch.float32 and not x.requires_grad
             and self.kernel_size == (8, 8) and self.stride == (4, 4) and self.padding == (0, 0)
-            and (x.shape[-1] - 8) // 4 + 1 <= 32
+            and (x.shape[-1] - 8) // 4 + 1 in (15, 20)
         ):
             from ding.ops import dispatch as _dispatch
             if _dispatch.use_hip_autograd(x):
