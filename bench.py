@@ -122,22 +122,42 @@ def build_impala(device: str, multi_gpu: bool):
     return model, optimizer
 
 
+_IMPALA_GRAPH = {}
+
+
 def impala_step(model, optimizer, device: str, batch_size: int = 128, unroll_len: int = 32):
     from ding.rl_utils import vtrace_data, vtrace_error_discrete_action
     T, B = unroll_len, batch_size
     obs = torch.rand(T + 1, B, 4, 84, 84, device=device)
-    flat = obs.view((T + 1) * B, 4, 84, 84)
-    out = model(flat, mode='compute_actor_critic')
-    logit = out['logit'].view(T + 1, B, -1)
-    value = out['value'].view(T + 1, B)
     action = torch.randint(0, 6, (T, B), device=device)
     reward = torch.randn(T, B, device=device)
-    behaviour = logit[:-1].detach() + 0.1 * torch.randn_like(logit[:-1])
-    data = vtrace_data(logit[:-1], behaviour, action, value, reward, None)
-    loss = vtrace_error_discrete_action(data, gamma=0.99, lambda_=0.95)
-    total = loss.policy_loss + 0.5 * loss.value_loss - 0.01 * loss.entropy_loss
-    optimizer.zero_grad()
-    total.backward()
+
+    def step_fn(inp):
+        flat = inp['obs'].view((T + 1) * B, 4, 84, 84)
+        out = model(flat, mode='compute_actor_critic')
+        logit = out['logit'].view(T + 1, B, -1)
+        value = out['value'].view(T + 1, B)
+        behaviour = logit[:-1].detach() + 0.1 * torch.randn_like(logit[:-1])
+        data = vtrace_data(logit[:-1], behaviour, inp['action'], value, inp['reward'], None)
+        loss = vtrace_error_discrete_action(data, gamma=0.99, lambda_=0.95)
+        total = loss.policy_loss + 0.5 * loss.value_loss - 0.01 * loss.entropy_loss
+        optimizer.zero_grad(set_to_none=False)
+        total.backward()
+        return {'total': total.detach()}
+
+    graph_ok = (
+        device != 'cpu' and not torch.distributed.is_initialized()
+        and os.environ.get('DING_IMPALA_GRAPH', '1') not in ('0', 'false')
+    )
+    if graph_ok:
+        # hipGraph-captured fwd+v-trace+bwd (launch-bound at [33, 128])
+        if 'g' not in _IMPALA_GRAPH:
+            from ding.torch_utils.hip_graph import GraphedStep
+            _IMPALA_GRAPH['g'] = GraphedStep(step_fn)
+        _IMPALA_GRAPH['g']({'obs': obs, 'action': action, 'reward': reward})
+        optimizer.step()
+        return T * B
+    step_fn({'obs': obs, 'action': action, 'reward': reward})
     if torch.distributed.is_initialized():
         from ding.parallel import sync_gradients_flat
         sync_gradients_flat(model)

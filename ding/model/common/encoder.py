@@ -60,7 +60,7 @@ class StemConv2d(nn.Conv2d):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         import os
         if (
-            os.environ.get('DING_STEM_CONV', '1') not in ('0', 'false')
+            os.environ.get('DING_STEM_CONV', '0') in ('1', 'true')
             and x.is_cuda and x.dtype == torch.float32 and not x.requires_grad
             and self.kernel_size == (8, 8) and self.stride == (4, 4) and self.padding == (0, 0)
             and (x.shape[-1] - 8) // 4 + 1 in (15, 20)

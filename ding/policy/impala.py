@@ -71,6 +71,10 @@ class IMPALAPolicy(Policy):
         self._rho_clip_ratio = self._cfg.learn.rho_clip_ratio
         self._c_clip_ratio = self._cfg.learn.c_clip_ratio
         self._rho_pg_clip_ratio = self._cfg.learn.rho_pg_clip_ratio
+        # hipGraph capture of the whole learn step (fwd + v-trace + bwd):
+        # static [T, B] shapes make IMPALA a one-graph-per-step workload
+        self._cuda_graph = self._cfg.learn.get('cuda_graph', False) and not self._cfg.multi_gpu
+        self._graphed_step = None
         self._learn_model.reset()
 
     def _data_preprocess_learn(self, data: List[Dict[str, Any]]):
@@ -82,38 +86,55 @@ class IMPALAPolicy(Policy):
         data['obs_plus_1'] = torch.cat([data['obs'], data['next_obs'][-1:]], dim=0).float()
         return data
 
-    def _forward_learn(self, data: List[Dict[str, Any]]) -> Dict[str, Any]:
-        data = self._data_preprocess_learn(data)
-        self._learn_model.train()
+    def _learn_step(self, data: Dict[str, Any]) -> Dict[str, Any]:
+        """fwd + v-trace + bwd on already-device-resident [T(+1), B] tensors;
+        sync-free (returns 0-dim GPU tensors) so it can be hipGraph-captured."""
         T, B = data['done'].shape[:2]
         obs_flat = data['obs_plus_1'].reshape(-1, *data['obs_plus_1'].shape[2:])
         output = self._learn_model.forward(obs_flat, mode='compute_actor_critic')
         target_logit = output['logit'].reshape(T + 1, B, -1)[:-1]
         value = output['value'].reshape(T + 1, B)
-        behaviour_logit = data['logit']  # [T, B, N]
-        actions = data['action']  # [T, B]
-        rewards = data['reward']  # [T, B]
+        rewards = data['reward']
         if rewards.dim() == 3:
             rewards = rewards.squeeze(-1)
-        weights = data['weight']
-        # mask value after done: vtrace assumes within-trajectory; unroll slices
-        vt_data = vtrace_data(target_logit, behaviour_logit, actions, value, rewards, weights)
+        vt_data = vtrace_data(target_logit, data['logit'], data['action'], value, rewards, data.get('weight'))
         loss = vtrace_error_discrete_action(
             vt_data, self._gamma, self._lambda, self._rho_clip_ratio, self._c_clip_ratio, self._rho_pg_clip_ratio
         )
         total_loss = loss.policy_loss + self._value_weight * loss.value_loss \
             - self._entropy_weight * loss.entropy_loss
-        self._optimizer.zero_grad()
+        self._optimizer.zero_grad(set_to_none=False)
         total_loss.backward()
+        return {
+            'total_loss': total_loss.detach(),
+            'policy_loss': loss.policy_loss.detach(),
+            'value_loss': loss.value_loss.detach(),
+            'entropy_loss': loss.entropy_loss.detach(),
+        }
+
+    def _forward_learn(self, data: List[Dict[str, Any]]) -> Dict[str, Any]:
+        data = self._data_preprocess_learn(data)
+        self._learn_model.train()
+        use_graph = (
+            self._cuda_graph and isinstance(data.get('obs_plus_1'), torch.Tensor)
+            and data['obs_plus_1'].is_cuda and data.get('weight') is None
+        )
+        if use_graph:
+            if self._graphed_step is None:
+                from ding.torch_utils.hip_graph import GraphedStep
+                self._graphed_step = GraphedStep(self._learn_step)
+            inputs = {k: v for k, v in data.items() if isinstance(v, torch.Tensor)}
+            out = self._graphed_step(inputs)
+            self._optimizer.step()
+            losses = {k: float(v) for k, v in out.items()}
+            return {'cur_lr': self._optimizer.defaults['lr'], **losses}
+        out = self._learn_step(data)
         if self._cfg.multi_gpu:
             self.sync_gradients(self._model)
         self._optimizer.step()
         return {
             'cur_lr': self._optimizer.defaults['lr'],
-            'total_loss': total_loss.item(),
-            'policy_loss': loss.policy_loss.item(),
-            'value_loss': loss.value_loss.item(),
-            'entropy_loss': loss.entropy_loss.item(),
+            **{k: v.item() for k, v in out.items()},
         }
 
     def _monitor_vars_learn(self) -> List[str]:
