@@ -23,9 +23,11 @@ def compute_importance_weights(
     grad_ctx = torch.enable_grad() if requires_grad else torch.no_grad()
     with grad_ctx:
         if action_space_type == 'discrete':
-            dist_t = Categorical(logits=target_output)
-            dist_b = Categorical(logits=behaviour_output)
-            rhos = dist_t.log_prob(action) - dist_b.log_prob(action)
+            # explicit log-softmax (Categorical's arg validation hides a
+            # device->host sync, which also breaks hipGraph capture)
+            lp_t = torch.log_softmax(target_output, -1).gather(-1, action.long().unsqueeze(-1)).squeeze(-1)
+            lp_b = torch.log_softmax(behaviour_output, -1).gather(-1, action.long().unsqueeze(-1)).squeeze(-1)
+            rhos = lp_t - lp_b
         elif action_space_type == 'continuous':
             dist_t = Independent(Normal(target_output['mu'], target_output['sigma']), 1)
             dist_b = Independent(Normal(behaviour_output['mu'], behaviour_output['sigma']), 1)

@@ -64,10 +64,12 @@ def vtrace_error_discrete_action(
         adv = vtrace_advantage(pg_rhos, reward, return_t_plus_1, value[:-1], gamma)
     if weight is None:
         weight = torch.ones_like(reward)
-    dist_target = Categorical(logits=target_output)
-    pg_loss = -(dist_target.log_prob(action) * adv * weight).mean()
+    logp = F.log_softmax(target_output, -1)
+    logp_a = logp.gather(-1, action.long().unsqueeze(-1)).squeeze(-1)
+    pg_loss = -(logp_a * adv * weight).mean()
     value_loss = (F.mse_loss(value[:-1], return_, reduction='none') * weight).mean()
-    entropy_loss = (dist_target.entropy() * weight).mean()
+    entropy = -(logp.exp() * logp).sum(-1)
+    entropy_loss = (entropy * weight).mean()
     return vtrace_loss(pg_loss, value_loss, entropy_loss)
 
 
