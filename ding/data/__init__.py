@@ -7,3 +7,4 @@ from .storage import Storage, FileStorage
 from .storage_loader import StorageLoader, FileStorageLoader
 from .model_loader import ModelLoader, FileModelLoader
 from .level_replay import LevelSampler
+from .buffer import GPUPrioritizedBuffer

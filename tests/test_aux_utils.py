@@ -86,3 +86,31 @@ def test_cluster_script_generation(tmp_path):
     assert 'parallelism: 2' in manifest
     script = generate_slurm_script('job', 'python bench.py', nodes=2, output_path=str(tmp_path / 'job.sh'))
     assert '#SBATCH --nodes=2' in script
+
+
+def test_gpu_prioritized_buffer_cpu_semantics():
+    """GPUPrioritizedBuffer math runs on CPU device too (device='cpu'):
+    ring wrap, prioritized sampling bias, IS weights, priority updates."""
+    import torch
+    from ding.data import GPUPrioritizedBuffer
+    buf = GPUPrioritizedBuffer(size=64, alpha=1.0, beta=1.0, device='cpu')
+    batch = {'obs': torch.randn(32, 4), 'reward': torch.randn(32)}
+    idx = buf.push(batch)
+    assert buf.count() == 32 and idx.shape == (32, )
+    # second push wraps the ring
+    buf.push({'obs': torch.randn(48, 4), 'reward': torch.randn(48)})
+    assert buf.count() == 64
+    # bias: give one index huge priority, it should dominate samples
+    hot = torch.tensor([5])
+    buf.update_priority(hot, torch.tensor([1000.0]))
+    sampled, sidx, w = buf.sample(256)
+    assert sampled['obs'].shape == (256, 4)
+    frac_hot = (sidx == 5).float().mean()
+    assert frac_hot > 0.5, float(frac_hot)
+    assert w.max() <= 1.0 + 1e-6
+    # hot samples get the smallest IS weight
+    assert torch.isclose(w[sidx == 5].max(), w.min())
+    sd = buf.state_dict()
+    buf2 = GPUPrioritizedBuffer(size=64, device='cpu')
+    buf2.load_state_dict(sd)
+    assert buf2.count() == 64
