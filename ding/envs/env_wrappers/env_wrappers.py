@@ -418,6 +418,7 @@ class RamWrapper(EnvWrapper):
         return np.asarray(obs, dtype=np.float32).reshape(-1, 1, 1), reward, done, info
 
 
+@ENV_WRAPPER_REGISTRY.register('obs_plus_prev_action_reward')
 @ENV_WRAPPER_REGISTRY.register('obs_plus_prev_act_rew')
 class ObsPlusPrevActRewWrapper(EnvWrapper):
     """NGU input contract: obs dict {obs, prev_action, prev_reward_extrinsic}
@@ -484,6 +485,7 @@ class GymToGymnasiumWrapper(EnvWrapper):
         return out
 
 
+@ENV_WRAPPER_REGISTRY.register('reward_in_obs')
 @ENV_WRAPPER_REGISTRY.register('all_in_obs')
 class AllinObsWrapper(EnvWrapper):
     """Decision-Transformer input contract: obs dict {obs, reward}

@@ -12,7 +12,7 @@ import torch
 
 from ding.envs import BaseEnvManager
 from ding.torch_utils import to_ndarray, to_tensor
-from ding.utils import EasyDict, SERIAL_COLLECTOR_REGISTRY, build_logger, deep_merge_dicts
+from ding.utils import EasyDict, SERIAL_COLLECTOR_REGISTRY, SERIAL_EVALUATOR_REGISTRY, build_logger, deep_merge_dicts
 
 from .sample_serial_collector import ISerialCollector
 
@@ -253,3 +253,88 @@ class BattleEpisodeSerialCollector(BattleSampleSerialCollector):
                         return_info[p].append(info)
 
         return return_data, return_info
+
+
+@SERIAL_EVALUATOR_REGISTRY.register('battle_interaction')
+class BattleInteractionSerialEvaluator:
+    """Evaluate policy 0 of a battle env against fixed opponents; stop on
+    player-0 mean return reaching stop_value.
+
+    Parity: reference ding/worker/collector/
+    battle_interaction_serial_evaluator.py ('battle_interaction':17).
+    """
+
+    config = dict(type='battle_interaction', eval_freq=50, n_episode=4, stop_value=1e9)
+
+    @classmethod
+    def default_config(cls) -> EasyDict:
+        return EasyDict(cls.config)
+
+    def __init__(self, cfg, env=None, policy=None, tb_logger=None,
+                 exp_name: str = 'default_experiment', instance_name: str = 'battle_evaluator'):
+        self._cfg = deep_merge_dicts(self.default_config(), cfg or EasyDict({}))
+        self._env = env
+        self._policy = list(policy)
+        self._policy_num = len(self._policy)
+        self._stop_value = self._cfg.stop_value
+        self._last_eval_iter = -1
+
+    def reset(self, _policy=None, _env=None) -> None:
+        if _env is not None:
+            self._env = _env
+        if _policy is not None:
+            self._policy = list(_policy)
+        for p in self._policy:
+            p.reset()
+
+    def close(self) -> None:
+        if self._env is not None:
+            self._env.close()
+
+    def should_eval(self, train_iter: int) -> bool:
+        if train_iter == self._last_eval_iter:
+            return False
+        if (train_iter - self._last_eval_iter) < self._cfg.eval_freq and train_iter != 0:
+            return False
+        self._last_eval_iter = train_iter
+        return True
+
+    def eval(self, save_ckpt_fn=None, train_iter: int = -1, envstep: int = -1, n_episode=None):
+        import numpy as np
+        n_episode = n_episode or self._cfg.n_episode
+        if self._env.closed:
+            self._env.launch()
+        else:
+            self._env.reset()
+        for p in self._policy:
+            p.reset()
+        returns = []
+        while len(returns) < n_episode:
+            obs = self._env.ready_obs
+            if not isinstance(obs, dict) or not all(isinstance(k, int) for k in obs):
+                ids = self._env.ready_obs_id
+                obs = {i: obs[pos] for pos, i in enumerate(ids)}
+            obs = {i: to_tensor(o, dtype=torch.float32) for i, o in obs.items()}
+            per_policy_obs = [
+                {env_id: o[p] for env_id, o in obs.items()} for p in range(self._policy_num)
+            ]
+            with torch.no_grad():
+                outs = [pol.forward(per_policy_obs[p]) for p, pol in enumerate(self._policy)]
+            actions = {}
+            for p, out_d in enumerate(outs):
+                for env_id, out in out_d.items():
+                    actions.setdefault(env_id, []).append(to_ndarray(out['action']))
+            timesteps = self._env.step(actions)
+            if not isinstance(timesteps, dict):
+                timesteps = {ts.info['env_id']: ts for ts in timesteps}
+            for env_id, ts in timesteps.items():
+                if ts.done:
+                    info0 = ts.info[0] if isinstance(ts.info, (list, tuple)) else ts.info
+                    returns.append(float(info0.get('eval_episode_return', 0.0)))
+                    for p in self._policy:
+                        p.reset([env_id])
+        mean_ret = float(np.mean(returns))
+        stop = mean_ret >= self._stop_value and train_iter > 0
+        if stop and save_ckpt_fn:
+            save_ckpt_fn('ckpt_best.pth.tar')
+        return stop, {'eval_episode_return': returns, 'train_iter': train_iter}

@@ -18,9 +18,10 @@ from typing import Any, Callable, Dict, List, Optional, Union
 import torch
 
 from ding.interaction import Slave, TaskFail
-from ding.utils import EasyDict
+from ding.utils import COMM_COLLECTOR_REGISTRY, COMM_LEARNER_REGISTRY, EasyDict
 
 
+@COMM_LEARNER_REGISTRY.register('flask_fs')
 class FlaskFileSystemLearner(Slave):
     """Learner worker: receives {learner_start_task, learner_get_data_task,
     learner_learn_task, learner_close_task}; data referenced by FS paths."""
@@ -103,6 +104,7 @@ class FlaskFileSystemLearner(Slave):
         return out
 
 
+@COMM_COLLECTOR_REGISTRY.register('flask_fs')
 class FlaskFileSystemCollector(Slave):
     """Collector worker: receives {collector_start_task, collector_data_task,
     collector_close_task}; writes collected samples under path_data and

@@ -282,3 +282,41 @@ class SequenceReplayBuffer(NaiveReplayBuffer):
                         continue  # physical wrap inside window: skip
                 starts.append(s)
             return [[self._data[(s + j) % n] for j in range(sequence)] for s in starts]
+
+
+@BUFFER_REGISTRY.register('elastic')
+class ElasticReplayBuffer(NaiveReplayBuffer):
+    """Naive buffer whose SAMPLEABLE window grows on a schedule:
+    cfg.set_buffer_size(envstep) -> current window size (newest-first).
+
+    Parity: reference ding/worker/replay_buffer/naive_buffer.py
+    ElasticReplayBuffer ('elastic':381).
+    """
+
+    config = dict(
+        type='elastic',
+        replay_buffer_size=10000,
+        deepcopy=False,
+        enable_track_used_data=False,
+        periodic_thruput_seconds=60,
+        set_buffer_size=lambda envstep: 10000,
+    )
+
+    def __init__(self, cfg, tb_logger=None, exp_name: str = 'default_experiment', instance_name: str = 'buffer'):
+        super().__init__(cfg, tb_logger, exp_name, instance_name)
+        self._set_buffer_size = self._cfg.set_buffer_size
+        self._current_buffer_size = self._set_buffer_size(0)
+
+    def update_buffer_size(self, envstep: int) -> None:
+        self._current_buffer_size = self._set_buffer_size(envstep)
+
+    def sample(self, size: int, cur_learner_iter: int = -1, sample_range=None):
+        if size == 0:
+            return []
+        with self._lock:
+            window = min(self._valid_count, self._current_buffer_size)
+            if window < size:
+                return None
+            # newest `window` entries of the ring
+            idx = (self._tail - 1 - np.random.choice(window, size, replace=False)) % self._replay_buffer_size
+            return [self._data[i] for i in idx]

@@ -146,3 +146,23 @@ def test_metric_serial_evaluator():
     assert ev.should_eval(1)
     stop, result = ev.eval(None, train_iter=1)
     assert stop and result['acc'] == 1.0
+
+
+def test_battle_interaction_evaluator():
+    from ding.worker import BattleInteractionSerialEvaluator
+    pols = [_FlattenBattlePolicy(p) for p in _two_dqn_policies()]
+    env = _battle_env_manager()
+    ev = BattleInteractionSerialEvaluator(
+        EasyDict({'type': 'battle_interaction', 'n_episode': 3, 'stop_value': 1e9, 'eval_freq': 1}),
+        env=env, policy=pols, exp_name='exp/test_battle_eval'
+    )
+    assert ev.should_eval(1)
+    # DQN collect_mode needs eps; wrap to default it
+    class _Eps:
+        def __init__(self, p): self._p = p
+        def forward(self, obs, **kw): return self._p.forward(obs, eps=0.1)
+        def reset(self, *a, **k): return self._p.reset(*a, **k)
+    ev.reset(_policy=[_Eps(p) for p in pols])
+    stop, info = ev.eval(None, train_iter=1)
+    assert not stop and len(info['eval_episode_return']) >= 3
+    ev.close()
