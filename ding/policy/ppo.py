@@ -111,7 +111,25 @@ class PPOPolicy(Policy):
         # by graphs, so multi_gpu keeps the eager path.
         self._cuda_graph = self._cfg.learn.get('cuda_graph', False) and not self._cfg.multi_gpu
         self._graphed_step = None
+        self._values_graph = None
         self._learn_model.reset()
+
+    def _graphed_values(self, both: torch.Tensor, chunk: int) -> torch.Tensor:
+        """hipGraph-captured no-grad chunked critic pass for recompute-adv."""
+        if getattr(self, '_values_graph', None) is None:
+            from ding.torch_utils.hip_graph import GraphedStep
+
+            def fn(inp):
+                with torch.no_grad():
+                    chunks = torch.split(inp['both'], chunk, dim=0)
+                    return {
+                        'values': torch.cat(
+                            [self._learn_model.forward(c, mode='compute_critic')['value'] for c in chunks], dim=0
+                        )
+                    }
+
+            self._values_graph = GraphedStep(fn)
+        return self._values_graph({'both': both})['values'].clone()
 
     def _graphed_minibatch(self, batch: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
         """Replay (or first capture) the fwd+loss+bwd hipGraph for one
@@ -167,18 +185,22 @@ class PPOPolicy(Policy):
                 with torch.no_grad():
                     if isinstance(data['obs'], torch.Tensor):
                         # one batched critic pass over [obs; next_obs], chunked
-                        # to <=3200 rows: MIOpen's tuned conv solvers cover the
-                        # minibatch-family shapes; huge batches fall to naive
-                        # fp64-accum kernels (seen in rocprof on MI355X)
+                        # at the learn minibatch size: those conv shapes are
+                        # already tuned (rocprof: batch-3200 critic passes fell
+                        # to MIOpen's naive fp64-accum convs at ~17 ms each).
+                        # Under cuda_graph the whole chunked pass is captured
+                        # once and replayed per epoch (~200 eager launches ->
+                        # one hipGraphLaunch).
                         both = torch.cat([data['obs'], data['next_obs']], dim=0)
-                        # chunk at the learn minibatch size: those conv shapes
-                        # are already tuned (rocprof: batch-3200 critic passes
-                        # fell to naive fp64-accum convs at ~17 ms each)
                         _chunk = max(1, int(self._cfg.learn.batch_size))
-                        chunks = torch.split(both, _chunk, dim=0)
-                        values = torch.cat(
-                            [self._learn_model.forward(c, mode='compute_critic')['value'] for c in chunks], dim=0
-                        )
+                        if self._cuda_graph and both.is_cuda and both.dtype == torch.float32:
+                            values = self._graphed_values(both, _chunk)
+                        else:
+                            chunks = torch.split(both, _chunk, dim=0)
+                            values = torch.cat(
+                                [self._learn_model.forward(c, mode='compute_critic')['value'] for c in chunks],
+                                dim=0
+                            )
                         value, next_value = values.chunk(2, dim=0)
                     else:
                         value = self._learn_model.forward(data['obs'], mode='compute_critic')['value']
