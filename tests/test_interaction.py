@@ -113,3 +113,32 @@ def test_comm_collector_fs_roundtrip(tmp_path):
     data = torch.load(tmp_path / out['data_path'], weights_only=False)
     assert len(data) == 8 and 'obs' in data[0]
     worker._process_task({'name': 'collector_close_task'})
+
+
+def test_dist_entry_in_process(tmp_path):
+    """Legacy distributed loop: commander -> collector task -> FS data ->
+    learner task -> policy saved -> next collector cycle reloads it."""
+    from ding.entry.dist_entry import dist_launch_coordinator
+    from ding.worker.comm import FlaskFileSystemCollector, FlaskFileSystemLearner
+    from ding.utils import EasyDict
+    from ding.config import compile_config
+    from tests.test_policy_breadth import cartpole_cfg
+
+    main, create = cartpole_cfg('dqn', extra_policy=dict(nstep=1))
+    create.policy.type = 'dqn_command'
+    cfg = compile_config(main, create_cfg=create, auto=True, save_cfg=False)
+    cfg.policy.type = 'dqn'
+
+    comm_cfg = EasyDict({'path_data': str(tmp_path), 'path_policy': str(tmp_path)})
+    learner = FlaskFileSystemLearner(comm_cfg)
+    collector = FlaskFileSystemCollector(comm_cfg)
+    coord_cfg = EasyDict(dict(commander=dict(
+        type='solo', eval_interval=int(1e9),
+        collector_cfg=dict(cfg=cfg, n_sample=16),
+        policy=dict(cfg.policy),
+        policy_id='p.pth',
+    )))
+    hist = dist_launch_coordinator(coord_cfg, learner=learner, collector=collector, max_cycles=3)
+    assert len(hist['collect']) >= 1
+    assert len(hist['learn']) >= 1
+    assert hist['learn'][0]['train_iter'] >= 1
