@@ -1,0 +1,166 @@
+"""GPU one-step learn smokes across policy families: each builds a tiny
+policy with cuda=True, runs one _forward_learn on synthetic data, and
+checks finite losses. Complements tests/test_ops_gpu.py's kernel numerics —
+these exercise the full policy paths (fused kernels + MIOpen/hipBLASLt) on
+the MI355X."""
+import numpy as np
+import pytest
+import torch
+
+from ding.utils import EasyDict, deep_merge_dicts
+
+pytestmark = pytest.mark.gpu
+
+
+def _mk(policy_type, extra=None, enable=('learn', )):
+    from ding.policy import create_policy
+    from ding.utils.registry import POLICY_REGISTRY
+    cls = POLICY_REGISTRY.get(policy_type)
+    cfg = EasyDict(deep_merge_dicts(cls.default_config(), EasyDict(dict(type=policy_type, cuda=True, **(extra or {})))))
+    return create_policy(cfg, enable_field=list(enable))
+
+
+def _trans(n=16, obs=8, act=3, discrete=True, nstep=1):
+    out = []
+    for _ in range(n):
+        d = {
+            'obs': torch.randn(obs), 'next_obs': torch.randn(obs),
+            'action': torch.tensor(np.random.randint(act)) if discrete else torch.randn(act),
+            'reward': torch.randn(nstep) if nstep > 1 else torch.randn(1),
+            'done': False, 'collect_iter': 0,
+        }
+        out.append(d)
+    return out
+
+
+def test_dqn_gpu_learn():
+    pol = _mk('dqn', dict(nstep=3, model=dict(obs_shape=8, action_shape=3, encoder_hidden_size_list=[32, 32]),
+                          learn=dict(batch_size=16)))
+    info = pol._forward_learn(_trans(nstep=3))
+    assert np.isfinite(info['total_loss'])
+
+
+def test_r2d2_gpu_learn():
+    """R2D2 exercises the fused LN-LSTM cell + n-step rescale TD on GPU."""
+    from ding.utils.data import timestep_collate
+    pol = _mk('r2d2', dict(
+        model=dict(obs_shape=8, action_shape=3, encoder_hidden_size_list=[32, 32], lstm_type='normal'),
+        learn=dict(batch_size=4), collect=dict(unroll_len=8), unroll_len=8, nstep=2, burnin_step=2,
+    ))
+    samples = []
+    for _ in range(4):
+        T = 8
+        samples.append({
+            'obs': [torch.randn(8) for _ in range(T)],
+            'action': [torch.tensor(np.random.randint(3)) for _ in range(T)],
+            'reward': [torch.randn(2) for _ in range(T)],
+            'done': [False] * T,
+            'prev_state': [None] * T,
+            'weight': None,
+        })
+    info = pol._forward_learn(samples)
+    assert np.isfinite(info['total_loss'])
+
+
+def test_sac_gpu_learn():
+    pol = _mk('sac', dict(
+        model=dict(obs_shape=8, action_shape=3, twin_critic=True, action_space='reparameterization'),
+        learn=dict(batch_size=16, auto_alpha=True),
+    ))
+    info = pol._forward_learn(_trans(discrete=False))
+    assert np.isfinite(info['total_loss'])
+
+
+def test_c51_gpu_learn():
+    """C51 exercises the HIP categorical-projection kernel on GPU."""
+    pol = _mk('c51', dict(nstep=3, model=dict(obs_shape=8, action_shape=3, encoder_hidden_size_list=[32, 32],
+                                              v_min=-10, v_max=10, n_atom=51),
+                          learn=dict(batch_size=16)))
+    info = pol._forward_learn(_trans(nstep=3))
+    assert np.isfinite(info['total_loss'])
+
+
+def test_qmix_gpu_learn():
+    from ding.utils.data import timestep_collate
+    pol = _mk('qmix', dict(
+        model=dict(agent_num=3, obs_shape=6, global_obs_shape=10, action_shape=4,
+                   hidden_size_list=[16, 16, 16]),
+        learn=dict(batch_size=4), collect=dict(unroll_len=5),
+    ))
+    T, A = 5, 3
+    samples = []
+    for _ in range(4):
+        samples.append({
+            'obs': [
+                {'agent_state': torch.randn(A, 6), 'global_state': torch.randn(10),
+                 'action_mask': torch.ones(A, 4)} for _ in range(T)
+            ],
+            'next_obs': [
+                {'agent_state': torch.randn(A, 6), 'global_state': torch.randn(10),
+                 'action_mask': torch.ones(A, 4)} for _ in range(T)
+            ],
+            'action': [torch.randint(0, 4, (A, )) for _ in range(T)],
+            'reward': [torch.randn(1) for _ in range(T)],
+            'done': [False] * T,
+            'prev_state': [None] * T,
+            'weight': None,
+        })
+    info = pol._forward_learn(samples)
+    assert np.isfinite(info['total_loss'])
+
+
+def test_dreamer_gpu_learn():
+    """DreamerV3 on GPU: RSSM world-model train + latent-imagination policy."""
+    from ding.world_model import create_world_model
+    from ding.worker import create_buffer
+    wm_cfg = EasyDict(dict(
+        type='dreamer', import_names=['ding.world_model.dreamer'],
+        train_freq=1, eval_freq=int(1e9), cuda=True,
+        model=dict(state_size=4, obs_type='vector', action_size=2, action_type='discrete',
+                   encoder_hidden_size_list=[32, 32], dyn_stoch=8, dyn_deter=32, dyn_hidden=32,
+                   dyn_discrete=8, units=32, reward_layers=1, discount_layers=1, image_dec_layers=1,
+                   batch_size=4, batch_length=6),
+    ))
+    wm = create_world_model(wm_cfg)
+    buf = create_buffer(EasyDict({'type': 'sequence', 'replay_buffer_size': 1000}))
+    for i in range(64):
+        buf.push({'obs': np.random.randn(4).astype(np.float32), 'action': np.int64(i % 2),
+                  'reward': np.float32(0.1), 'done': False})
+    post, _ = wm.train(buf, envstep=10, train_iter=0, batch_size=4, batch_length=6)
+    pol = _mk('dreamer', dict(
+        imag_horizon=4,
+        model=dict(action_shape=2, dyn_stoch=8, dyn_deter=32, dyn_discrete=8, units=32,
+                   actor_layers=1, value_layers=1, actor_dist='onehot'),
+        learn=dict(batch_size=4, batch_length=6),
+        collect=dict(unroll_len=1, action_size=2, collect_dyn_sample=True),
+    ))
+    info = pol._forward_learn(post, world_model=wm, envstep=10)
+    assert np.isfinite(info['actor_loss']) and np.isfinite(info['critic_loss'])
+
+
+def test_diffuser_gpu_learn():
+    from ding.model.template.diffusion import PlanDiffuser
+    m = PlanDiffuser(
+        diffuser_model='GaussianDiffusion',
+        diffuser_model_cfg=dict(model='DiffusionUNet1d',
+                                model_cfg=dict(transition_dim=6, dim=16, dim_mults=[1, 2]),
+                                horizon=8, obs_dim=4, action_dim=2, n_timesteps=8, clip_denoised=True),
+        value_model=None, value_model_cfg=None,
+    ).cuda()
+    x = torch.randn(6, 8, 6, device='cuda')
+    cond = {0: torch.randn(6, 4, device='cuda')}
+    t = torch.randint(0, 8, (6, ), device='cuda')
+    loss, _ = m.diffuser_loss(x, cond, t)
+    loss.backward()
+    assert torch.isfinite(loss)
+
+
+def test_gpu_prioritized_buffer_device():
+    from ding.data import GPUPrioritizedBuffer
+    buf = GPUPrioritizedBuffer(size=128, device='cuda')
+    buf.push({'obs': torch.randn(64, 8), 'reward': torch.randn(64)})
+    batch, idx, w = buf.sample(32)
+    assert batch['obs'].is_cuda and idx.is_cuda and w.is_cuda
+    buf.update_priority(idx, torch.rand(32, device='cuda') * 10)
+    batch2, _, _ = buf.sample(32)
+    assert batch2['obs'].shape == (32, 8)
