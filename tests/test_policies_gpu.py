@@ -56,7 +56,6 @@ def test_r2d2_gpu_learn():
             'reward': [torch.randn(2) for _ in range(T)],
             'done': [False] * T,
             'prev_state': [None] * T,
-            'weight': None,
         })
     info = pol._forward_learn(samples)
     assert np.isfinite(info['total_loss'])
@@ -103,7 +102,6 @@ def test_qmix_gpu_learn():
             'reward': [torch.randn(1) for _ in range(T)],
             'done': [False] * T,
             'prev_state': [None] * T,
-            'weight': None,
         })
     info = pol._forward_learn(samples)
     assert np.isfinite(info['total_loss'])
