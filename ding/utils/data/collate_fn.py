@@ -89,7 +89,14 @@ def timestep_collate(batch: List[Dict[str, Any]]) -> Dict[str, Any]:
 
     out = {}
     for key in elem:
-        out[key] = _stack_time([d[key] for d in batch])
+        vals = [d[key] for d in batch]
+        # an optional per-sample field (e.g. weight=None) collapses to None
+        if all(v is None for v in vals) or (
+            isinstance(vals[0], (list, tuple)) and all(x is None for v in vals for x in v)
+        ):
+            out[key] = None
+            continue
+        out[key] = _stack_time(vals)
     if prev_state is not None:
         out['prev_state'] = [list(s) for s in prev_state]
     return out
