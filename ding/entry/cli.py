@@ -49,7 +49,14 @@ def cli(mode, config, seed, env, policy, train_iter, env_step, load_path, replay
         from .parallel_entry import parallel_pipeline
         parallel_pipeline(config, seed)
     elif mode == 'dist':
-        raise click.UsageError('dist mode: use `ditask` with --main and topology flags')
+        # legacy role launcher: --platform slurm/k8s fills per-process args;
+        # the modern multi-process path is `ditask`
+        from ding.entry.dist_entry import dist_prepare_config
+        click.echo(
+            'dist mode (legacy): use ding.entry.dist_entry launchers '
+            '(dist_launch_coordinator/learner/collector) or `ditask` for the event-bus runtime'
+        )
+        return
     else:
         raise click.UsageError(f'unimplemented mode {mode}')
 
