@@ -192,3 +192,57 @@ def test_ppof_policy():
     assert ev['action'].shape == (5, )
     sd = pol.state_dict()
     pol.load_state_dict(sd)
+
+
+def test_minigrid_lite_rnd_pipeline():
+    """Sparse-reward gridworld + RND intrinsic reward end-to-end."""
+    from ding.entry import serial_pipeline_reward_model
+    main = EasyDict(dict(
+        exp_name='exp/test_minigrid_rnd',
+        env=dict(collector_env_num=2, evaluator_env_num=2, n_evaluator_episode=2, stop_value=2,
+                 grid_size=5, max_step=30),
+        policy=dict(
+            cuda=False, nstep=1, discount_factor=0.99,
+            model=dict(obs_shape=5 * 5 * 4 + 4, action_shape=3, encoder_hidden_size_list=[32, 32]),
+            learn=dict(update_per_collect=2, batch_size=16, learning_rate=1e-3),
+            collect=dict(n_sample=16),
+            eval=dict(evaluator=dict(eval_freq=int(1e6))),
+            other=dict(eps=dict(type='exp', start=0.95, end=0.1, decay=10000),
+                       replay_buffer=dict(replay_buffer_size=1000)),
+        ),
+        reward_model=dict(type='rnd', obs_shape=5 * 5 * 4 + 4, hidden_size_list=[16, 16],
+                          update_per_collect=1),
+    ))
+    create = EasyDict(dict(
+        env=dict(type='minigrid_lite', import_names=['dizoo.gridworld.envs.minigrid_lite_env']),
+        env_manager=dict(type='base'),
+        policy=dict(type='dqn'),
+    ))
+    serial_pipeline_reward_model((main, create), seed=0, max_train_iter=2)
+
+
+def test_memory_len_r2d2_pipeline():
+    """Memory env + R2D2 (recurrent unrolls through the fused-LSTM path on
+    GPU; eager here) for 2 iterations."""
+    from ding.entry import serial_pipeline
+    main = EasyDict(dict(
+        exp_name='exp/test_memory_r2d2',
+        env=dict(collector_env_num=2, evaluator_env_num=2, n_evaluator_episode=2, stop_value=0.99,
+                 memory_length=6),
+        policy=dict(
+            cuda=False, priority=True, priority_IS_weight=True,
+            model=dict(obs_shape=3, action_shape=2, encoder_hidden_size_list=[16, 16], lstm_type='normal'),
+            discount_factor=0.99, nstep=2, burnin_step=1, unroll_len=6, learn_unroll_len=5,
+            learn=dict(update_per_collect=2, batch_size=4, learning_rate=1e-3, target_update_theta=0.01),
+            collect=dict(n_sample=8, unroll_len=6, env_num=2),
+            eval=dict(evaluator=dict(eval_freq=int(1e6)), env_num=2),
+            other=dict(eps=dict(type='exp', start=0.95, end=0.1, decay=10000),
+                       replay_buffer=dict(replay_buffer_size=1000)),
+        ),
+    ))
+    create = EasyDict(dict(
+        env=dict(type='memory_len', import_names=['dizoo.memory.envs.memory_len_env']),
+        env_manager=dict(type='base'),
+        policy=dict(type='r2d2'),
+    ))
+    serial_pipeline((main, create), seed=0, max_train_iter=2)
