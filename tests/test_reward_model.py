@@ -107,3 +107,43 @@ def test_ngu():
 def test_factory():
     rm = create_reward_model(EasyDict(dict(type='rnd', obs_shape=4, hidden_size_list=[16, 16])))
     assert isinstance(rm, RndRewardModel)
+
+
+def test_her_bitflip_relabel():
+    """HER on BitFlip: relabeling with 'final' strategy turns a failed
+    episode's last transition into a success (reward 1)."""
+    import torch
+    from ding.reward_model import HerRewardModel
+    from dizoo.bitflip.envs import BitFlipEnv
+
+    env = BitFlipEnv({'n_bits': 6})
+    env.seed(3, dynamic_seed=False)
+    obs = env.reset()
+    episode = []
+    rng = __import__('numpy').random.RandomState(0)
+    for _ in range(12):
+        action = int(rng.randint(0, 6))
+        ts = env.step(action)
+        episode.append({
+            'obs': torch.as_tensor(obs), 'next_obs': torch.as_tensor(ts.obs),
+            'action': torch.tensor([action]), 'reward': torch.as_tensor(ts.reward), 'done': ts.done,
+        })
+        obs = ts.obs
+        if ts.done:
+            break
+
+    def goal_fn(t):
+        # achieved goal = the state bits of next_obs
+        return t['next_obs'][:6]
+
+    def reward_fn(goal, t):
+        ok = torch.allclose(t['next_obs'][:6].float(), torch.as_tensor(goal).float())
+        return torch.ones_like(t['reward']) if ok else torch.zeros_like(t['reward'])
+
+    her = HerRewardModel({'her_strategy': 'final', 'her_replay_k': 2,
+                          'goal_fn': goal_fn, 'reward_fn': reward_fn})
+    relabelled = her.estimate(episode)
+    assert len(relabelled) == 2
+    # final-strategy: last transition always achieves the (relabelled) goal
+    for ep in relabelled:
+        assert float(ep[-1]['reward']) == 1.0
