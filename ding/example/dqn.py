@@ -1,0 +1,35 @@
+"""CartPole DQN through the Task middleware pipeline (reference
+ding/example/dqn.py)."""
+from ding.data import DequeBuffer
+from ding.framework import OnlineRLContext, task
+from ding.framework.middleware import (
+    CkptSaver, OffPolicyLearner, StepCollector, data_pusher, eps_greedy_handler, interaction_evaluator,
+    nstep_reward_enhancer, termination_checker,
+)
+from ding.policy import DQNPolicy
+from .common import cartpole_envs, compile
+
+
+def main(max_step: int = 1000, exp_name: str = 'exp/example_dqn'):
+    from dizoo.classic_control.cartpole.config.cartpole_dqn_config import create_config, main_config
+    cfg = compile(main_config, create_config, exp_name)
+    collector_env, evaluator_env = cartpole_envs(cfg)
+    policy = DQNPolicy(cfg.policy)
+    buffer_ = DequeBuffer(size=cfg.policy.other.replay_buffer.replay_buffer_size)
+    with task.start(ctx=OnlineRLContext()):
+        task.use(interaction_evaluator(cfg, policy.eval_mode, evaluator_env))
+        task.use(eps_greedy_handler(cfg))
+        task.use(StepCollector(cfg, policy.collect_mode, collector_env))
+        task.use(nstep_reward_enhancer(cfg))
+        task.use(data_pusher(cfg, buffer_))
+        task.use(OffPolicyLearner(cfg, policy.learn_mode, buffer_))
+        task.use(CkptSaver(policy, cfg.exp_name, train_freq=1000))
+        task.use(termination_checker(max_env_step=int(1e5)))
+        task.run(max_step=max_step)
+    collector_env.close()
+    evaluator_env.close()
+    return policy
+
+
+if __name__ == '__main__':
+    main()

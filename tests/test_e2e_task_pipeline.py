@@ -120,3 +120,11 @@ def test_step_timer_wrapper():
         pass
     assert calls == ['fwd', 'bwd']
     assert len(timer.records['mw']) == 1
+
+
+@pytest.mark.parametrize('mod', ['dqn', 'dqn_per', 'ppo', 'sac'])
+def test_example_mains(mod, tmp_path):
+    """ding/example/* mains run a few pipeline steps end-to-end."""
+    import importlib
+    m = importlib.import_module(f'ding.example.{mod}')
+    m.main(max_step=6, exp_name=str(tmp_path / mod))
