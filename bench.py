@@ -84,7 +84,10 @@ def make_ppo_batch(n_sample: int, device: str, policy) -> dict:
     B = n_sample
     logit = out['logit'].detach().repeat((B + 319) // 320, 1)[:B].contiguous()
     value = out['value'].detach().repeat((B + 319) // 320)[:B].contiguous()
-    action = torch.distributions.Categorical(logits=logit).sample()
+    # gumbel-max sample: sync-free (Categorical's arg validation forces a
+    # device->host sync per step)
+    gumbel = -torch.log(-torch.log(torch.rand_like(logit) + 1e-10) + 1e-10)
+    action = (logit + gumbel).argmax(dim=-1)
     reward = torch.sign(torch.randn(B, device=device)) * (torch.rand(B, device=device) < 0.05)
     done = (torch.rand(B, device=device) < 0.002).float()
     return {
