@@ -10,7 +10,7 @@ The ring is allocated lazily from the first pushed batch's schema; pushes,
 samples and priority updates are all batched tensor ops on the buffer's
 device, so a learner can sample minibatches without leaving the GPU.
 """
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, Optional, Tuple
 
 import torch
 

@@ -4,7 +4,7 @@ error) mixed with a staleness term.
 
 Parity: reference ding/data/level_replay/level_sampler.py (LevelSampler:9).
 """
-from typing import List, Optional, Union
+from typing import List, Optional
 
 import numpy as np
 import torch

@@ -9,7 +9,6 @@ interaction/comm stack: the coordinator drives a parallel commander's task
 loop directly against in-process (or HTTP-attached) comm workers; the
 modern path for new code remains the event-bus Task runtime (`ditask`).
 """
-import os
 import time
 from typing import Any, Dict, Optional
 

@@ -5,11 +5,9 @@ Parity: reference ding/model/template/diffusion.py (GaussianDiffusion:73,
 ValueDiffusion:296, PlanDiffuser 'pd':321, GaussianInvDynDiffusion 'dd':372,
 default_sample_fn:15, n_step_guided_p_sample:37).
 """
-import math
 from collections import namedtuple
-from typing import Optional, Union
+from typing import Optional
 
-import numpy as np
 import torch
 import torch.nn as nn
 import torch.nn.functional as F

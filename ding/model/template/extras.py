@@ -9,7 +9,7 @@ bcq.py ('bcq':13), ngu.py ('ngu':44), procedure_cloning.py
 qvac.py ('continuous_qvac':13).
 """
 import copy
-from typing import Dict, List, Optional, Union
+from typing import Dict, List, Optional
 
 import torch
 import torch.nn as nn

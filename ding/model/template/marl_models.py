@@ -10,14 +10,14 @@ for the QTRAN-base losses, CollaQ decomposes per-agent Q into a self part
 plus an ally-attention part.
 """
 import copy
-from typing import Dict, List, Optional, Sequence, Union
+from typing import List, Optional
 
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
 from ding.utils import MODEL_REGISTRY
-from .qmix import QMix, Mixer
+from .qmix import QMix
 from .mavac import MAVAC
 
 

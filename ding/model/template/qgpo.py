@@ -9,8 +9,6 @@ compact second-order DPM-Solver++ for the linear VP schedule implemented
 here; everything stays on-device (no .cpu() round-trips in sample()).
 """
 import copy
-from typing import Optional
-
 import torch
 import torch.nn as nn
 import torch.nn.functional as F

@@ -6,7 +6,6 @@ VAC-style module exposing compute_actor/compute_critic/compute_actor_critic,
 and the discrete learn path reuses the fused HIP PPO loss via ppo_error.
 """
 import copy
-import random
 from collections import namedtuple
 from typing import Any, Callable, Dict, List, Optional
 

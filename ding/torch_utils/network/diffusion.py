@@ -14,7 +14,6 @@ from typing import List, Optional
 
 import torch
 import torch.nn as nn
-import torch.nn.functional as F
 
 
 def extract(a: torch.Tensor, t: torch.Tensor, x_shape) -> torch.Tensor:

@@ -9,7 +9,7 @@ Re-designed: plain-Python math, fp32-first, no device strings threaded
 through constructors (modules follow `.to(device)` like the rest of ding).
 """
 import math
-from typing import Callable, List, Optional, Tuple
+from typing import Callable, List, Tuple
 
 import numpy as np
 import torch

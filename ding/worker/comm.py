@@ -10,10 +10,8 @@ Re-designed around our compact interaction.Slave: task verbs arrive as
 ``path_policy``.
 """
 import os
-import threading
-import time
 import uuid
-from typing import Any, Callable, Dict, List, Optional, Union
+from typing import Any, List, Union
 
 import torch
 

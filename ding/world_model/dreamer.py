@@ -19,7 +19,6 @@ from ding.torch_utils.network.dreamer import (
     ContDist, DenseHead, GRUCellLN, OneHotDist, static_scan, weight_init, uniform_weight_init,
 )
 from ding.utils import WORLD_MODEL_REGISTRY, EasyDict, deep_merge_dicts
-from ding.utils.data import default_collate
 from .base_world_model import WorldModel
 
 
