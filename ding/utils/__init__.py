@@ -37,3 +37,5 @@ from .loader import Loader, LoaderError, is_type, to_type, interval, enum, item,
 from .misc_helpers import (
     get_vi_sequence, render, K8sLauncher, generate_slurm_script, find_free_port, node_to_partition, node_to_host,
 )
+from .misc_helpers import get_ip, get_pid, PropagatingThread, deprecated
+from typing import Sequence as SequenceType  # reference utils/type_helper.py

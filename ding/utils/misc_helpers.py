@@ -5,6 +5,7 @@ Parity: reference ding/utils/bfs_helper.py, render_helper.py,
 k8s_helper.py:118, slurm_helper.py, orchestrator_launcher.py:7.
 """
 import os
+import threading
 from typing import Any, List, Optional, Tuple
 
 import numpy as np
@@ -132,3 +133,59 @@ def node_to_partition(node: str) -> str:
 
 def node_to_host(node: str) -> str:
     return node
+
+
+def get_ip() -> str:
+    """Best-effort non-loopback IP (reference utils/system_helper.py:10)."""
+    import socket
+    try:
+        s = socket.socket(socket.AF_INET, socket.SOCK_DGRAM)
+        s.connect(('10.255.255.255', 1))
+        ip = s.getsockname()[0]
+        s.close()
+        return ip
+    except Exception:
+        return '127.0.0.1'
+
+
+def get_pid() -> int:
+    import os
+    return os.getpid()
+
+
+class PropagatingThread(threading.Thread):
+    """Thread that re-raises its exception in join() (reference
+    utils/system_helper.py:40)."""
+
+    def run(self):
+        self._exc = None
+        try:
+            self._ret = self._target(*self._args, **self._kwargs)
+        except Exception as e:
+            self._exc = e
+
+    def join(self, timeout=None):
+        super().join(timeout)
+        if getattr(self, '_exc', None) is not None:
+            raise self._exc
+        return getattr(self, '_ret', None)
+
+
+def deprecated(since: str, removed_in: str, up_to: str = None):
+    """Mark an API deprecated; warns on call (reference utils/deprecation.py)."""
+    import functools
+    import warnings
+
+    def deco(fn):
+
+        @functools.wraps(fn)
+        def wrapper(*args, **kwargs):
+            msg = f'{fn.__name__} is deprecated since {since}, will be removed in {removed_in}'
+            if up_to:
+                msg += f'; use {up_to} instead'
+            warnings.warn(msg, DeprecationWarning, stacklevel=2)
+            return fn(*args, **kwargs)
+
+        return wrapper
+
+    return deco
