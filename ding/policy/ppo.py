@@ -115,7 +115,10 @@ class PPOPolicy(Policy):
         self._learn_model.reset()
 
     def _graphed_values(self, both: torch.Tensor, chunk: int) -> torch.Tensor:
-        """hipGraph-captured no-grad chunked critic pass for recompute-adv."""
+        """hipGraph-captured no-grad chunked critic pass for recompute-adv.
+        The [obs; next_obs] tensor is identical across the epoch loop, so the
+        722 MB static-input copy happens once per _forward_learn (tracked by
+        tensor identity), later epochs replay copy-free."""
         if getattr(self, '_values_graph', None) is None:
             from ding.torch_utils.hip_graph import GraphedStep
 
@@ -129,7 +132,15 @@ class PPOPolicy(Policy):
                     }
 
             self._values_graph = GraphedStep(fn)
-        return self._values_graph({'both': both})['values'].clone()
+            self._values_graph_src = None
+        g = self._values_graph
+        key = g._shape_key({'both': both})
+        if g._graph is not None and key == g._key and self._values_graph_src == id(both):
+            g._graph.replay()
+            return g._static_out['values'].clone()
+        out = g({'both': both})['values'].clone()
+        self._values_graph_src = id(both)
+        return out
 
     def _graphed_minibatch(self, batch: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
         """Replay (or first capture) the fwd+loss+bwd hipGraph for one
@@ -180,6 +191,7 @@ class PPOPolicy(Policy):
             data['next_obs'] = data['next_obs'].float()
         self._learn_model.train()
         return_infos = []
+        _both_cache = None
         for epoch in range(self._cfg.learn.epoch_per_collect):
             if self._recompute_adv:
                 with torch.no_grad():
@@ -191,7 +203,9 @@ class PPOPolicy(Policy):
                         # Under cuda_graph the whole chunked pass is captured
                         # once and replayed per epoch (~200 eager launches ->
                         # one hipGraphLaunch).
-                        both = torch.cat([data['obs'], data['next_obs']], dim=0)
+                        if _both_cache is None:
+                            _both_cache = torch.cat([data['obs'], data['next_obs']], dim=0)
+                        both = _both_cache
                         _chunk = max(1, int(self._cfg.learn.batch_size))
                         if self._cuda_graph and both.is_cuda and both.dtype == torch.float32:
                             values = self._graphed_values(both, _chunk)
@@ -234,7 +248,9 @@ class PPOPolicy(Policy):
                     and isinstance(batch['obs'], torch.Tensor) and batch['obs'].is_cuda
                     and batch.get('weight') is None and batch['obs'].dtype == torch.float32
                 ):
-                    out = self._graphed_minibatch(batch)
+                    out = self._graphed_minibatch({
+                        k: batch[k] for k in ('obs', 'logit', 'action', 'value', 'adv', 'return')
+                    })
                     # static outputs: clone (async) now, convert to floats once
                     # at the end of the epoch — avoids a device sync per minibatch
                     graph_infos.append({k: v.clone() for k, v in out.items()})
