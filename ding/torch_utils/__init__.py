@@ -12,3 +12,4 @@ from .loss import LabelSmoothCELoss, SoftLogitsLoss, MultiLogitsLoss, Contrastiv
 from .math_helper import cov, unsqueeze_repeat
 from .reshape_helper import fold_batch, unfold_batch
 from .lr_scheduler import cos_lr_scheduler, get_lr_ratio
+from .extras import enable_tf32, get_num_params, levenshtein_distance, hamming_distance, is_differentiable, NonegativeParameter, TanhParameter, CategoricalPd, CategoricalPdPytorch, DataParallel
