@@ -114,3 +114,23 @@ def test_gpu_prioritized_buffer_cpu_semantics():
     buf2 = GPUPrioritizedBuffer(size=64, device='cpu')
     buf2.load_state_dict(sd)
     assert buf2.count() == 64
+
+
+@pytest.mark.benchmark
+def test_buffer_throughput_benchmark():
+    """Buffer micro-benchmark (SURVEY §4 benchmark-marked buffer test):
+    push/sample throughput of DequeBuffer; loose sanity floor only."""
+    import time
+    import torch
+    from ding.data import DequeBuffer
+    buf = DequeBuffer(size=10000)
+    data = [{'obs': torch.randn(8), 'action': 1, 'reward': 0.5} for _ in range(1000)]
+    t0 = time.time()
+    for d in data * 5:
+        buf.push(d)
+    push_rate = 5000 / (time.time() - t0)
+    t0 = time.time()
+    for _ in range(50):
+        buf.sample(256)
+    sample_rate = 50 * 256 / (time.time() - t0)
+    assert push_rate > 1e3 and sample_rate > 1e3, (push_rate, sample_rate)
