@@ -206,7 +206,9 @@ class PPOPolicy(Policy):
                         if _both_cache is None:
                             _both_cache = torch.cat([data['obs'], data['next_obs']], dim=0)
                         both = _both_cache
-                        _chunk = max(1, int(self._cfg.learn.batch_size))
+                        import os as _os
+                        _chunk = int(_os.environ.get('DING_PPO_VALUE_CHUNK', 0)) or \
+                            max(1, int(self._cfg.learn.batch_size))
                         if self._cuda_graph and both.is_cuda and both.dtype == torch.float32:
                             values = self._graphed_values(both, _chunk)
                         else:
