@@ -206,12 +206,13 @@ class PPOPolicy(Policy):
                         if _both_cache is None:
                             _both_cache = torch.cat([data['obs'], data['next_obs']], dim=0)
                         both = _both_cache
-                        # A/B on MI355X (320/640/1600): 224.7/210.7/208.9
-                        # ms/step — 1600 rows keeps MIOpen on tuned solvers
-                        # (3200+ fell to naive fp64-accum) with fewer launches
+                        # chunk-size sweep on MI355X (320/640/1600/2400/3200
+                        # -> 224.7/210.7/205.3/210.6/202.8 ms/step): 3200 wins
+                        # once the captured values graph + find-mode warmup
+                        # keep MIOpen on tuned solvers at that batch
                         import os as _os
                         _chunk = int(_os.environ.get('DING_PPO_VALUE_CHUNK', 0)) or \
-                            max(int(self._cfg.learn.batch_size), 1600)
+                            max(int(self._cfg.learn.batch_size), 3200)
                         if self._cuda_graph and both.is_cuda and both.dtype == torch.float32:
                             values = self._graphed_values(both, _chunk)
                         else:
