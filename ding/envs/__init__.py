@@ -15,3 +15,4 @@ from .env_manager.subprocess_env_manager import (
 )
 from .common.spaces import Discrete, Box, MultiDiscrete
 from .env_manager.env_supervisor import EnvSupervisor
+from .env.default_wrapper import get_default_wrappers
