@@ -1,5 +1,5 @@
 from .functional import *  # noqa
-from .collector import StepCollector, EpisodeCollector
+from .collector import PPOFStepCollector, StepCollector, EpisodeCollector
 from .learner import OffPolicyLearner, HERLearner
 from .ckpt_handler import CkptSaver
 from .distributer import ContextExchanger, ModelExchanger, PeriodicalModelExchanger

@@ -94,3 +94,43 @@ def get_random_policy(cfg: EasyDict, policy: Policy, env: BaseEnvManager):
     if cfg.policy.get('random_collect', None) and cfg.policy.random_collect.get('use_policy', False):
         return policy
     return _RandomView()
+
+
+class PPOFStepCollector:
+    """Step collector for the simplified PPOF policy interface (policy.collect
+    on stacked obs tensors; transitions assembled via policy.process_transition).
+
+    Parity: reference ding/framework/middleware/collector.py PPOFStepCollector.
+    """
+
+    def __init__(self, seed: int, policy, env: BaseEnvManager, n_sample: int, unroll_len: int = 1) -> None:
+        self._policy = policy
+        self._env = env
+        self._n_sample = n_sample
+        if env.closed:
+            env.launch()
+        self._transitions = []
+
+    def __call__(self, ctx) -> None:
+        import torch
+        collected = []
+        while len(collected) < self._n_sample:
+            obs = self._env.ready_obs
+            if not isinstance(obs, dict) or not all(isinstance(k, int) for k in obs):
+                ids = self._env.ready_obs_id
+                obs = {i: obs[pos] for pos, i in enumerate(ids)}
+            ids = sorted(obs.keys())
+            stacked = torch.stack([torch.as_tensor(obs[i], dtype=torch.float32) for i in ids])
+            out = self._policy.collect(stacked)
+            actions = {i: out['action'][k].cpu().numpy() for k, i in enumerate(ids)}
+            timesteps = self._env.step(actions)
+            if not isinstance(timesteps, dict):
+                timesteps = {ts.info['env_id']: ts for ts in timesteps}
+            for k, i in enumerate(ids):
+                if i not in timesteps:
+                    continue
+                ts = timesteps[i]
+                per = {kk: (vv[k] if isinstance(vv, torch.Tensor) else vv) for kk, vv in out.items()}
+                collected.append(self._policy.process_transition(stacked[k], per, ts))
+        ctx.trajectories = collected
+        ctx.env_step = getattr(ctx, 'env_step', 0) + len(collected)

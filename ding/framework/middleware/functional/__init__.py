@@ -11,3 +11,5 @@ from .enhancer import reward_estimator, her_data_enhancer, nstep_reward_enhancer
 from .termination_checker import termination_checker, ddp_termination_checker, epoch_timer, final_ctx_saver
 from .logger import online_logger, offline_logger, wandb_online_logger, wandb_offline_logger
 from .priority import priority_calculator
+from .data_processor import buffer_saver as offline_data_saver  # reference-name alias
+from .data_processor import offline_data_fetcher_from_mem as OfflineMemoryDataFetcher  # reference-name alias
