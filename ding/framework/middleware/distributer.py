@@ -47,8 +47,11 @@ class ContextExchanger:
         if payload:
             if self._storage_loader and task.has_role(task.role.COLLECTOR):
                 payload = self._storage_loader.save(payload)
-            for role in task.roles:
-                task.emit(self._event_name.format(role=role), payload, only_remote=True)
+            # address the PEER roles' topics (a node listens on its own
+            # role topic; emitting to our own topic would loop back to us)
+            for role in task.role:
+                if not task.has_role(role):
+                    task.emit(self._event_name.format(role=role), payload, only_remote=True)
 
     def __del__(self):
         if self._storage_loader is not None:
@@ -189,15 +192,10 @@ class ModelExchanger:
             self._update_model()
 
     def _update_model(self):
-        start = time.time()
-        while True:
-            if task.finish or self._state_dict_cache is not None:
-                break
-            if time.time() - start > 60:
-                logger.warning("timeout waiting for model update")
-                return
-            time.sleep(0.01)
-        if task.finish:
+        # non-blocking: run with the current (possibly stale) weights until
+        # the learner's broadcast lands — blocking here deadlocks against the
+        # learner, which waits for trajectories before its first send
+        if self._state_dict_cache is None or task.finish:
             return
         sd = self._state_dict_cache
         self._state_dict_cache = None

@@ -178,3 +178,53 @@ def test_parallel_auto_recover():
         os.remove(_CRASH_FILE)
     Parallel.runner(n_parallel_workers=1, auto_recover=True, max_retries=1)(_flaky_main)
     os.remove(_CRASH_FILE)
+
+
+def _exchanger_main():
+    """2 nodes: node 0 = learner, node 1 = collector. The collector produces
+    trajectories + env_step; ContextExchanger ships them to the learner; the
+    learner bumps train_iter which flows back. ModelExchanger broadcasts the
+    learner's weights to the collector."""
+    import time
+    import torch
+    import torch.nn as nn
+    from ding.framework import OnlineRLContext, Role
+    from ding.framework import task as _task
+    from ding.framework.middleware import ContextExchanger, ModelExchanger
+
+    model = nn.Linear(4, 2)
+    with _task.start(ctx=OnlineRLContext()):
+        is_learner = _task.router.node_id == 0
+        _task.add_role(Role.LEARNER if is_learner else Role.COLLECTOR)
+        _task.use(ContextExchanger(skip_n_iter=1))
+        _task.use(ModelExchanger(model))
+
+        if is_learner:
+            with torch.no_grad():
+                for p in model.parameters():
+                    p.fill_(0.5)
+
+            def learner_mw(ctx):
+                if ctx.total_step >= 1:
+                    assert ctx.trajectories is not None and len(ctx.trajectories) == 4, ctx.trajectories
+                    assert ctx.env_step > 0
+                ctx.train_iter += 1
+
+            _task.use(learner_mw)
+        else:
+
+            def collector_mw(ctx):
+                ctx.trajectories = [{'obs': torch.randn(4)} for _ in range(4)]
+                ctx.env_step = (ctx.env_step or 0) + 4
+
+            _task.use(collector_mw)
+        _task.run(max_step=5)
+        if not is_learner:
+            # ModelExchanger must have delivered the learner's 0.5-filled weights
+            assert all(torch.allclose(p, torch.full_like(p, 0.5)) for p in model.parameters()), \
+                [p.data for p in model.parameters()]
+
+
+def test_context_model_exchanger():
+    from ding.framework import Parallel
+    Parallel.runner(n_parallel_workers=2, topology='mesh', protocol='tcp', startup_interval=0.2)(_exchanger_main)
