@@ -228,3 +228,54 @@ def _exchanger_main():
 def test_context_model_exchanger():
     from ding.framework import Parallel
     Parallel.runner(n_parallel_workers=2, topology='mesh', protocol='tcp', startup_interval=0.2)(_exchanger_main)
+
+
+def _ditask_rl_main():
+    """Real distributed RL: node 0 learns, node 1 collects CartPole steps;
+    trajectories/model flow through Context/ModelExchanger over TCP."""
+    import torch
+    from ding.config import compile_config
+    from ding.data import DequeBuffer
+    from ding.envs import BaseEnvManagerV2
+    from ding.framework import OnlineRLContext, Role
+    from ding.framework import task as _task
+    from ding.framework.middleware import (
+        ContextExchanger, ModelExchanger, OffPolicyLearner, StepCollector, data_pusher, eps_greedy_handler,
+        termination_checker,
+    )
+    from ding.policy import DQNPolicy
+    from ding.utils import EasyDict
+    from dizoo.classic_control.cartpole.config.cartpole_dqn_config import create_config, main_config
+    from dizoo.classic_control.cartpole.envs.cartpole_env import CartPoleEnv
+
+    cfg = compile_config(main_config, create_cfg=create_config, auto=True, save_cfg=False, seed=0)
+    cfg.exp_name = 'exp/test_ditask_rl'
+    cfg.policy.learn.batch_size = 16
+    cfg.policy.collect.n_sample = 32
+    policy = DQNPolicy(cfg.policy)
+    with _task.start(ctx=OnlineRLContext()):
+        is_learner = _task.router.node_id == 0
+        _task.add_role(Role.LEARNER if is_learner else Role.COLLECTOR)
+        _task.use(ContextExchanger(skip_n_iter=1))
+        _task.use(ModelExchanger(policy._model))
+        if is_learner:
+            buffer_ = DequeBuffer(size=cfg.policy.other.replay_buffer.replay_buffer_size)
+            _task.use(data_pusher(cfg, buffer_))
+            _task.use(OffPolicyLearner(cfg, policy.learn_mode, buffer_))
+        else:
+            env = BaseEnvManagerV2(env_fn=[lambda: CartPoleEnv({}) for _ in range(2)], cfg=cfg.env.manager)
+            env.seed(_task.router.node_id)
+            _task.use(eps_greedy_handler(cfg))
+            _task.use(StepCollector(cfg, policy.collect_mode, env))
+        _task.use(termination_checker(max_env_step=int(1e5)))
+        _task.run(max_step=8)
+        if is_learner:
+            assert _task.ctx.train_iter > 0, "learner never trained"
+        else:
+            assert _task.ctx.env_step > 0
+            assert _task.ctx.train_iter > 0, "train_iter never flowed back to the collector"
+
+
+def test_ditask_actor_learner_rl():
+    from ding.framework import Parallel
+    Parallel.runner(n_parallel_workers=2, topology='mesh', protocol='tcp', startup_interval=0.2)(_ditask_rl_main)
