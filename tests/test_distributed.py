@@ -279,3 +279,28 @@ def _ditask_rl_main():
 def test_ditask_actor_learner_rl():
     from ding.framework import Parallel
     Parallel.runner(n_parallel_workers=2, topology='mesh', protocol='tcp', startup_interval=0.2)(_ditask_rl_main)
+
+
+def _barrier_main():
+    """Two nodes synchronize at a Barrier each iteration; both must complete
+    the same number of steps without deadlock."""
+    import time
+    from ding.framework import OnlineRLContext
+    from ding.framework import task as _task
+    from ding.framework.middleware import Barrier
+
+    with _task.start(ctx=OnlineRLContext()):
+        _task.use(Barrier(attch_from_nums=1))
+        steps = []
+
+        def mw(ctx):
+            steps.append(ctx.total_step)
+
+        _task.use(mw)
+        _task.run(max_step=4)
+        assert len(steps) == 4, steps
+
+
+def test_barrier_two_nodes():
+    from ding.framework import Parallel
+    Parallel.runner(n_parallel_workers=2, topology='mesh', protocol='tcp', startup_interval=0.2)(_barrier_main)
