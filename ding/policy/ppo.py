@@ -114,7 +114,7 @@ class PPOPolicy(Policy):
         self._values_graph = None
         self._learn_model.reset()
 
-    def _graphed_values(self, both: torch.Tensor, chunk: int) -> torch.Tensor:
+    def _graphed_values(self, both: torch.Tensor, chunk: int, fresh: bool = True) -> torch.Tensor:
         """hipGraph-captured no-grad chunked critic pass for recompute-adv.
         The [obs; next_obs] tensor is identical across the epoch loop, so the
         722 MB static-input copy happens once per _forward_learn (tracked by
@@ -132,15 +132,14 @@ class PPOPolicy(Policy):
                     }
 
             self._values_graph = GraphedStep(fn)
-            self._values_graph_src = None
         g = self._values_graph
         key = g._shape_key({'both': both})
-        if g._graph is not None and key == g._key and self._values_graph_src == id(both):
+        if not fresh and g._graph is not None and key == g._key:
+            # same tensor as the epoch-0 copy (caller tracks freshness —
+            # id() comparison would be unsafe across allocator reuse)
             g._graph.replay()
             return g._static_out['values'].clone()
-        out = g({'both': both})['values'].clone()
-        self._values_graph_src = id(both)
-        return out
+        return g({'both': both})['values'].clone()
 
     def _graphed_minibatch(self, batch: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
         """Replay (or first capture) the fwd+loss+bwd hipGraph for one
@@ -214,7 +213,7 @@ class PPOPolicy(Policy):
                         _chunk = int(_os.environ.get('DING_PPO_VALUE_CHUNK', 0)) or \
                             max(int(self._cfg.learn.batch_size), 3200)
                         if self._cuda_graph and both.is_cuda and both.dtype == torch.float32:
-                            values = self._graphed_values(both, _chunk)
+                            values = self._graphed_values(both, _chunk, fresh=(epoch == 0))
                         else:
                             chunks = torch.split(both, _chunk, dim=0)
                             values = torch.cat(
