@@ -34,6 +34,14 @@ import torch
 torch.backends.cudnn.benchmark = True
 
 
+def _maybe_channels_last(policy):
+    """Opt-in NHWC experiment (DING_CHANNELS_LAST=1): MIOpen picks different
+    solver families for NHWC fp32 convs on gfx950."""
+    if os.environ.get('DING_CHANNELS_LAST', '0') in ('1', 'true'):
+        policy._model.to(memory_format=torch.channels_last)
+    return policy
+
+
 def build_ppo_policy(device: str, multi_gpu: bool):
     from ding.policy import PPOPolicy
     from ding.utils import EasyDict, deep_merge_dicts
@@ -71,7 +79,7 @@ def build_ppo_policy(device: str, multi_gpu: bool):
         collect=dict(n_sample=3200, unroll_len=1, discount_factor=0.99, gae_lambda=0.95),
     )
     cfg = EasyDict(deep_merge_dicts(cfg, user))
-    return PPOPolicy(cfg, enable_field=['learn'])
+    return _maybe_channels_last(PPOPolicy(cfg, enable_field=['learn']))
 
 
 def make_ppo_batch(n_sample: int, device: str, policy) -> dict:
@@ -108,6 +116,9 @@ def ppo_step(policy, device: str, n_sample: int):
     # scale obs like the Atari pipeline (uint8 -> [0,1]) without a CPU trip
     data['obs'] = data['obs'].float().div_(255.0)
     data['next_obs'] = data['next_obs'].float().div_(255.0)
+    if os.environ.get('DING_CHANNELS_LAST', '0') in ('1', 'true'):
+        data['obs'] = data['obs'].to(memory_format=torch.channels_last)
+        data['next_obs'] = data['next_obs'].to(memory_format=torch.channels_last)
     policy._forward_learn(data)
     return n_sample
 
