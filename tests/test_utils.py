@@ -114,3 +114,60 @@ def test_env_supervisor():
     for _ in range(5):
         ts = sup.step({i: np.random.randint(2) for i in sup.ready_obs_id})
     sup.close()
+
+
+def test_new_env_wrappers():
+    """StaticObsNorm / Ram / Transpose / ObsPlusPrevActRew / AllinObs /
+    GymToGymnasium wrapper semantics."""
+    import numpy as np
+    from ding.envs.env_wrappers.env_wrappers import (
+        AllinObsWrapper, GymToGymnasiumWrapper, ObsPlusPrevActRewWrapper, RamWrapper, StaticObsNormWrapper,
+        TransposeWrapper,
+    )
+
+    class FakeEnv:
+
+        def __init__(self, obs):
+            self._obs = np.asarray(obs, dtype=np.float32)
+
+        def reset(self, **kw):
+            return self._obs
+
+        def step(self, action):
+            return self._obs, 1.0, False, {}
+
+    e = StaticObsNormWrapper(FakeEnv([10.0, 20.0]), mean=[10.0, 20.0], std=[2.0, 4.0])
+    assert np.allclose(e.reset(), [0, 0])
+    obs, r, d, i = e.step(0)
+    assert np.allclose(obs, [0, 0]) and r == 1.0
+
+    e = RamWrapper(FakeEnv(np.arange(8)))
+    assert e.reset().shape == (8, 1, 1)
+
+    e = TransposeWrapper(FakeEnv(np.zeros((4, 5, 3))))
+    assert e.reset().shape == (3, 4, 5)
+
+    e = ObsPlusPrevActRewWrapper(FakeEnv([1.0]))
+    o = e.reset()
+    assert o['prev_action'] == -1 and o['prev_reward_extrinsic'] == 0.0
+    o, _, _, _ = e.step(2)
+    o2, _, _, _ = e.step(3)
+    assert o2['prev_action'] == 2 and o2['prev_reward_extrinsic'] == 1.0
+
+    e = AllinObsWrapper(FakeEnv([1.0]))
+    o = e.reset()
+    assert set(o.keys()) == {'obs', 'reward'}
+
+    class GymnasiumEnv:
+
+        def reset(self, seed=None, **kw):
+            return np.zeros(2), {}
+
+        def step(self, action):
+            return np.zeros(2), 0.5, False, True, {}
+
+    e = GymToGymnasiumWrapper(GymnasiumEnv())
+    e.seed(3)
+    assert e.reset().shape == (2, )
+    obs, r, done, info = e.step(0)
+    assert done is True and r == 0.5  # truncated folds into done
