@@ -34,10 +34,14 @@ import torch
 torch.backends.cudnn.benchmark = True
 
 
+def _channels_last_on() -> bool:
+    # NHWC default: MIOpen's channels-last solvers measured 181.8 vs 211.5
+    # ms/step on the PPO bench (A/B bench_nhwc.json / bench_nchw.json)
+    return os.environ.get('DING_CHANNELS_LAST', '1') not in ('0', 'false')
+
+
 def _maybe_channels_last(policy):
-    """Opt-in NHWC experiment (DING_CHANNELS_LAST=1): MIOpen picks different
-    solver families for NHWC fp32 convs on gfx950."""
-    if os.environ.get('DING_CHANNELS_LAST', '0') in ('1', 'true'):
+    if _channels_last_on():
         policy._model.to(memory_format=torch.channels_last)
     return policy
 
@@ -116,7 +120,7 @@ def ppo_step(policy, device: str, n_sample: int):
     # scale obs like the Atari pipeline (uint8 -> [0,1]) without a CPU trip
     data['obs'] = data['obs'].float().div_(255.0)
     data['next_obs'] = data['next_obs'].float().div_(255.0)
-    if os.environ.get('DING_CHANNELS_LAST', '0') in ('1', 'true'):
+    if _channels_last_on():
         data['obs'] = data['obs'].to(memory_format=torch.channels_last)
         data['next_obs'] = data['next_obs'].to(memory_format=torch.channels_last)
     policy._forward_learn(data)
@@ -132,6 +136,8 @@ def build_impala(device: str, multi_gpu: bool):
         obs_shape=[4, 84, 84], action_shape=6, encoder_hidden_size_list=[128, 128, 256],
         actor_head_hidden_size=256, critic_head_hidden_size=256
     ).to(device)
+    if _channels_last_on():
+        model.to(memory_format=torch.channels_last)
     optimizer = Adam(model.parameters(), lr=6e-4, grad_clip_type='clip_norm', clip_value=5)
     return model, optimizer
 
@@ -143,6 +149,8 @@ def impala_step(model, optimizer, device: str, batch_size: int = 128, unroll_len
     from ding.rl_utils import vtrace_data, vtrace_error_discrete_action
     T, B = unroll_len, batch_size
     obs = torch.rand(T + 1, B, 4, 84, 84, device=device)
+    if _channels_last_on():
+        obs = obs.reshape((T + 1) * B, 4, 84, 84).to(memory_format=torch.channels_last).view(T + 1, B, 4, 84, 84)
     action = torch.randint(0, 6, (T, B), device=device)
     reward = torch.randn(T, B, device=device)
 
