@@ -121,9 +121,10 @@ class PPOPolicy(Policy):
 
     def _graphed_values(self, both: torch.Tensor, chunk: int, fresh: bool = True) -> torch.Tensor:
         """hipGraph-captured no-grad chunked critic pass for recompute-adv.
-        The [obs; next_obs] tensor is identical across the epoch loop, so the
-        722 MB static-input copy happens once per _forward_learn (tracked by
-        tensor identity), later epochs replay copy-free."""
+        The [obs; next_obs] tensor is identical across the epoch loop, so
+        the 722 MB static-input copy happens once per _forward_learn
+        (``fresh`` is True only on epoch 0); later epochs replay copy-free.
+        The chunk size is baked in at first capture."""
         if getattr(self, '_values_graph', None) is None:
             from ding.torch_utils.hip_graph import GraphedStep
 
