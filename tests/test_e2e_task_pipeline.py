@@ -122,7 +122,7 @@ def test_step_timer_wrapper():
     assert len(timer.records['mw']) == 1
 
 
-@pytest.mark.parametrize('mod', ['dqn', 'dqn_per', 'ppo', 'sac', 'dqn_rnd', 'ppo_offpolicy', 'cql', 'dqn_her'])
+@pytest.mark.parametrize('mod', ['dqn', 'dqn_per', 'ppo', 'sac', 'dqn_rnd', 'ppo_offpolicy', 'cql', 'dqn_her', 'dt'])
 def test_example_mains(mod, tmp_path):
     """ding/example/* mains run a few pipeline steps end-to-end."""
     import importlib
