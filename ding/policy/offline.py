@@ -286,7 +286,7 @@ class TD3BCPolicy(TD3Policy):
                 {'obs': collated['obs'], 'action': pred_action}, mode='compute_critic'
             )['q_value'][0]
             lam = self._alpha_bc / q.abs().mean().detach()
-            bc_loss = F.mse_loss(pred_action, collated['action'])
+            bc_loss = F.mse_loss(pred_action, collated['action'].reshape(pred_action.shape))
             actor_loss = -lam * q.mean() + bc_loss
             self._optimizer_actor.zero_grad()
             actor_loss.backward()
