@@ -297,7 +297,7 @@ class TD3BCPolicy(TD3Policy):
             'critic_loss': critic_loss.item(),
             'actor_loss': float(actor_loss.detach()),
             'bc_loss': float(bc_loss.detach()),
-            'total_loss': critic_loss.item() + float(actor_loss),
+            'total_loss': critic_loss.item() + float(actor_loss.detach()),
             'cur_lr': self._optimizer_critic.defaults['lr'],
         }
 
