@@ -73,7 +73,10 @@ def test_comm_learner_fs_roundtrip(tmp_path):
     assert 'cpu' in out
     out = learner_worker._process_task({
         'name': 'learner_start_task',
-        'task_info': {'policy': dict(cfg.policy, type='dqn'), 'learner_cfg': {}, 'policy_id': 'p.pth'},
+        'task_info': {
+            'policy': dict(cfg.policy, type='dqn'), 'learner_cfg': {}, 'policy_id': 'p.pth',
+            'exp_name': str(tmp_path / 'exp'),
+        },
     })
     assert 'started' in out['message']
     demand = learner_worker._process_task({'name': 'learner_get_data_task', 'task_id': 't0', 'buffer_id': 'b0'})
