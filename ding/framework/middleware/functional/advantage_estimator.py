@@ -19,8 +19,8 @@ def gae_estimator(cfg: EasyDict, policy: Policy, buffer_=None) -> Callable:
     """Compute GAE advantages over ctx.trajectories in one batched call; emit
     ctx.train_data (or push to buffer)."""
     model = policy.get_attribute('model')
-    # on-policy PPO with recompute_adv only needs raw fields + value/adv slots
-    rm_keys = ['next_obs'] if cfg.policy.get('recompute_adv', True) else []
+    # recompute_adv needs next_obs downstream; without it we can drop it to save memory
+    rm_keys = [] if cfg.policy.get('recompute_adv', True) else ['next_obs']
 
     def _gae(ctx: OnlineRLContext):
         data = ctx.trajectories  # list of transition dicts
@@ -46,7 +46,7 @@ def gae_estimator(cfg: EasyDict, policy: Policy, buffer_=None) -> Callable:
         )
         batch['adv'] = gae(adv_data, cfg.policy.collect.discount_factor, cfg.policy.collect.gae_lambda).squeeze(-1)
         for k in rm_keys:
-            pass  # keep next_obs: recompute_adv path needs it
+            batch.pop(k, None)
         if buffer_ is None:
             ctx.train_data = default_decollate(batch, ignore=['prev_state'])
             # keep as one collated dict for on-policy epoch training

@@ -11,9 +11,13 @@ sync_gradients) with an MI355X-native design:
 * In async mode the bucket all-reduce launches on a dedicated HIP stream as
   soon as the bucket's grads are ready, overlapping with the rest of
   backward; ``sync()`` waits and scatters the reduced flat buffer back.
-* The "participation indicator" semantics for partially-used networks are
-  preserved: params with no grad contribute zeros and a per-bucket
-  participation counter divides instead of world_size when requested.
+* With ``with_indicator=True`` the reference's partially-used-network
+  averaging (opendilab/DI-engine ding/policy/base_policy.py:415-460) is
+  reproduced: each bucket carries one extra indicator float per parameter
+  (1.0 if this rank produced a grad, else 0.0) appended to the flat buffer,
+  so the participation counts ride in the SAME all-reduce message as the
+  grads (no extra collective), and each parameter's reduced grad is divided
+  by the number of participating ranks instead of world_size.
 """
 from typing import Dict, List, Optional
 
@@ -28,10 +32,12 @@ class GradBucket:
         self.flat = flat
         self.offsets = offsets
         self.ready_count = 0
+        self.ready = False
         self.work = None
 
     def reset(self):
         self.ready_count = 0
+        self.ready = False
         self.work = None
 
 
@@ -52,6 +58,7 @@ class GradBucketAllReducer:
         self._buckets: List[GradBucket] = []
         self._param_to_bucket: Dict[int, GradBucket] = {}
         self._comm_stream = torch.cuda.Stream() if torch.cuda.is_available() else None
+        self._next_launch = 0
         self._build_buckets()
         if self.async_overlap:
             self._register_hooks()
@@ -67,7 +74,10 @@ class GradBucketAllReducer:
             if not current:
                 return
             device = current[0].device
-            flat = torch.zeros(current_numel, dtype=torch.float32, device=device)
+            # indicator mode: one trailing float per param carries this rank's
+            # participation bit inside the same all-reduce message
+            extra = len(current) if self.with_indicator else 0
+            flat = torch.zeros(current_numel + extra, dtype=torch.float32, device=device)
             offsets, off = [], 0
             for p in current:
                 offsets.append(off)
@@ -98,13 +108,17 @@ class GradBucketAllReducer:
     def _launch(self, bucket: GradBucket):
         world = dist.get_world_size()
         # pack grads into the flat buffer
-        for p, off in zip(bucket.params, bucket.offsets):
+        ind_off = bucket.offsets[-1] + bucket.params[-1].numel() if self.with_indicator else None
+        for j, (p, off) in enumerate(zip(bucket.params, bucket.offsets)):
             n = p.numel()
             if p.grad is not None:
                 bucket.flat[off:off + n].copy_(p.grad.detach().reshape(-1))
             else:
                 bucket.flat[off:off + n].zero_()
-        bucket.flat.div_(world)
+            if ind_off is not None:
+                bucket.flat[ind_off + j] = 1.0 if p.grad is not None else 0.0
+        if not self.with_indicator:
+            bucket.flat.div_(world)
         if self._comm_stream is not None:
             self._comm_stream.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(self._comm_stream):
@@ -118,28 +132,50 @@ class GradBucketAllReducer:
         bucket = self._param_to_bucket[id(param)]
         bucket.ready_count += 1
         if bucket.ready_count == len(bucket.params):
-            self._launch(bucket)
+            bucket.ready = True
+            # launch strictly in bucket order so every rank issues the same
+            # all-reduce sequence even when parameter usage differs per rank
+            self._drain_ready()
+
+    def _drain_ready(self):
+        while self._next_launch < len(self._buckets):
+            b = self._buckets[self._next_launch]
+            if not b.ready:
+                break
+            self._launch(b)
+            self._next_launch += 1
 
     def sync(self):
         """Finish outstanding bucket reductions and write back grads. In
         synchronous mode this launches all buckets now."""
         if not dist.is_available() or not dist.is_initialized():
             return
-        for bucket in self._buckets:
-            if bucket.work is None:
-                self._launch(bucket)
+        for bucket in self._buckets[self._next_launch:]:
+            self._launch(bucket)
+        self._next_launch = len(self._buckets)
         for bucket in self._buckets:
             bucket.work.wait()
         if self._comm_stream is not None:
             torch.cuda.current_stream().wait_stream(self._comm_stream)
         for bucket in self._buckets:
-            for p, off in zip(bucket.params, bucket.offsets):
+            counts = None
+            if self.with_indicator:
+                # divide each param's summed grad by its participation count
+                ind_off = bucket.offsets[-1] + bucket.params[-1].numel()
+                counts = bucket.flat[ind_off:ind_off + len(bucket.params)].clamp_min(1.0)
+                for j, (p, off) in enumerate(zip(bucket.params, bucket.offsets)):
+                    bucket.flat[off:off + p.numel()].div_(counts[j])
+                raw_counts = bucket.flat[ind_off:ind_off + len(bucket.params)].cpu()
+            for j, (p, off) in enumerate(zip(bucket.params, bucket.offsets)):
                 n = p.numel()
                 if p.grad is None:
+                    if counts is not None and raw_counts[j] == 0.0:
+                        continue  # no rank used this param: leave grad None
                     p.grad = bucket.flat[off:off + n].reshape(p.shape).clone()
                 else:
                     p.grad.detach().reshape(-1).copy_(bucket.flat[off:off + n])
             bucket.reset()
+        self._next_launch = 0
 
 
 def sync_gradients_flat(model: torch.nn.Module, bucket_bytes: int = 25 * 1024 * 1024):

@@ -126,7 +126,7 @@ class PDPolicy(Policy):
         t = torch.randint(0, self.n_timesteps, (batch_size, ), device=x.device).long()
         cond = data['conditions']
         loss_dict['diffuse_loss'], a0 = self._model.diffuser_loss(x, cond, t)
-        loss_dict['a0_loss'] = float(a0)
+        loss_dict['a0_loss'] = float(a0.detach()) if isinstance(a0, torch.Tensor) else float(a0)
         loss_dict['diffuse_loss'] = loss_dict['diffuse_loss'] / self.gradient_accumulate_every
         loss_dict['diffuse_loss'].backward()
         if self._forward_learn_cnt < self.value_step and self._model.value:

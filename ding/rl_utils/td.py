@@ -573,7 +573,9 @@ def _quantile_huber_loss(pred, target, taus, kappa: float = 1.0):
         huber = torch.where(diff.abs() <= kappa, 0.5 * diff.pow(2), kappa * (diff.abs() - 0.5 * kappa))
     else:
         huber = diff.abs()
-    rho = (taus.unsqueeze(2) - (diff.detach() < 0).float()).abs() * huber / max(kappa, 1.0)
+    rho = (taus.unsqueeze(2) - (diff.detach() < 0).float()).abs() * huber
+    if kappa > 0:
+        rho = rho / kappa
     return rho.sum(1).mean(1)
 
 

@@ -1,9 +1,11 @@
 """V-trace (IMPALA) losses, discrete and continuous.
 
 Parity: reference ding/rl_utils/vtrace.py (vtrace_nstep_return:9,
-vtrace_advantage:32, vtrace_error_discrete_action:73). On GPU the whole
-pipeline (IS ratios -> clipped reverse scan -> three-term loss) runs in the
-fused HIP kernels (ding/ops/csrc/vtrace_ops.hip).
+vtrace_advantage:32, vtrace_error_discrete_action:73). On GPU the clipped
+reverse scan dispatches to the HIP kernel in ding/ops/csrc/scan_ops.hip via
+ding/ops/dispatch.py; IS ratios and the three-term loss stay in PyTorch here —
+the fully fused path lives in vtrace_error_discrete_fused (vtrace_ops.hip)
+when available.
 """
 from collections import namedtuple
 
