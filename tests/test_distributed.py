@@ -81,6 +81,41 @@ def _body_grad_bucket_async(rank, world):
     return {'grad0': model[0].weight.grad.detach().numpy().tolist()}
 
 
+def _body_grad_bucket_indicator(rank, world):
+    """Partially-used network: rank 0 uses both heads, rank 1 only head_a.
+
+    With the participation indicator, head_a grads average over 2 ranks,
+    head_b grads over the 1 participating rank (not world_size).
+    """
+    from ding.parallel import GradBucketAllReducer
+
+    class TwoHead(torch.nn.Module):
+
+        def __init__(self):
+            super().__init__()
+            self.trunk = torch.nn.Linear(8, 16)
+            self.head_a = torch.nn.Linear(16, 4)
+            self.head_b = torch.nn.Linear(16, 4)
+
+    model = TwoHead()
+    reducer = GradBucketAllReducer(model, bucket_bytes=64, async_overlap=False, with_indicator=True)
+    reducer.broadcast_params(src=0)
+    x = torch.randn(5, 8) * (rank + 1)
+    h = torch.relu(model.trunk(x))
+    loss = model.head_a(h).pow(2).mean()
+    if rank == 0:
+        loss = loss + model.head_b(h).pow(2).mean()
+    loss.backward()
+    local = {n: (p.grad.clone() if p.grad is not None else None) for n, p in model.named_parameters()}
+    reducer.sync()
+    return {
+        'head_a_w': model.head_a.weight.grad.numpy().tolist(),
+        'head_b_w': model.head_b.weight.grad.numpy().tolist(),
+        'local_head_a_w': local['head_a.weight'].numpy().tolist(),
+        'local_head_b_w': None if local['head_b.weight'] is None else local['head_b.weight'].numpy().tolist(),
+    }
+
+
 def _body_policy_multi_gpu(rank, world):
     from ding.policy import DQNPolicy
     from ding.utils import EasyDict, deep_merge_dicts
@@ -121,6 +156,21 @@ def test_grad_bucket_sync_mode():
     assert np.allclose(g0, g1, atol=1e-6), "grads must match after all-reduce"
     l0, l1 = np.array(res[0]['local_grad0']), np.array(res[1]['local_grad0'])
     assert np.allclose(g0, (l0 + l1) / 2, atol=1e-5), "reduced grad must be the average"
+
+
+def test_grad_bucket_indicator_partial_use():
+    res = _run_dist('_body_grad_bucket_indicator')
+    import numpy as np
+    a0, a1 = np.array(res[0]['head_a_w']), np.array(res[1]['head_a_w'])
+    assert np.allclose(a0, a1, atol=1e-6)
+    la0, la1 = np.array(res[0]['local_head_a_w']), np.array(res[1]['local_head_a_w'])
+    assert np.allclose(a0, (la0 + la1) / 2, atol=1e-5), "used-by-all param averages over world"
+    # head_b only used on rank 0: averaged over 1 participant, i.e. rank0's local grad
+    b0, b1 = np.array(res[0]['head_b_w']), np.array(res[1]['head_b_w'])
+    lb0 = np.array(res[0]['local_head_b_w'])
+    assert res[1]['local_head_b_w'] is None
+    assert np.allclose(b0, lb0, atol=1e-6), "partially-used param divides by participants (1), not world"
+    assert np.allclose(b0, b1, atol=1e-6)
 
 
 def test_grad_bucket_async_mode():
