@@ -129,62 +129,64 @@ def ppo_step(policy, device: str, n_sample: int):
 
 
 def build_impala(device: str, multi_gpu: bool):
-    """IMPALA workload: model + optimizer + v-trace loss (policy class lands
-    in a later milestone; the learner math path is exercised directly)."""
-    from ding.model import VAC
-    from ding.torch_utils import Adam
-    model = VAC(
-        obs_shape=[4, 84, 84], action_shape=6, encoder_hidden_size_list=[128, 128, 256],
-        actor_head_hidden_size=256, critic_head_hidden_size=256
-    ).to(device)
-    if _channels_last_on():
-        model.to(memory_format=torch.channels_last)
-    optimizer = Adam(model.parameters(), lr=6e-4, grad_clip_type='clip_norm', clip_value=5)
-    return model, optimizer
+    """IMPALA workload through the real IMPALAPolicy (reference
+    spaceinvaders_impala_config.py:17-45: unroll_len=32, batch=128,
+    encoder [128,128,256])."""
+    from ding.policy import IMPALAPolicy
+    from ding.utils import EasyDict, deep_merge_dicts
 
-
-_IMPALA_GRAPH = {}
-
-
-def impala_step(model, optimizer, device: str, batch_size: int = 128, unroll_len: int = 32):
-    from ding.rl_utils import vtrace_data, vtrace_error_discrete_action
-    T, B = unroll_len, batch_size
-    obs = torch.rand(T + 1, B, 4, 84, 84, device=device)
-    if _channels_last_on():
-        obs = obs.reshape((T + 1) * B, 4, 84, 84).to(memory_format=torch.channels_last).view(T + 1, B, 4, 84, 84)
-    action = torch.randint(0, 6, (T, B), device=device)
-    reward = torch.randn(T, B, device=device)
-
-    def step_fn(inp):
-        flat = inp['obs'].view((T + 1) * B, 4, 84, 84)
-        out = model(flat, mode='compute_actor_critic')
-        logit = out['logit'].view(T + 1, B, -1)
-        value = out['value'].view(T + 1, B)
-        behaviour = logit[:-1].detach() + 0.1 * torch.randn_like(logit[:-1])
-        data = vtrace_data(logit[:-1], behaviour, inp['action'], value, inp['reward'], None)
-        loss = vtrace_error_discrete_action(data, gamma=0.99, lambda_=0.95)
-        total = loss.policy_loss + 0.5 * loss.value_loss - 0.01 * loss.entropy_loss
-        optimizer.zero_grad(set_to_none=False)
-        total.backward()
-        return {'total': total.detach()}
-
-    graph_ok = (
-        device != 'cpu' and not torch.distributed.is_initialized()
-        and os.environ.get('DING_IMPALA_GRAPH', '1') not in ('0', 'false')
+    cfg = IMPALAPolicy.default_config()
+    user = dict(
+        cuda=device.startswith('cuda'),
+        multi_gpu=multi_gpu,
+        unroll_len=32,
+        model=dict(
+            obs_shape=[4, 84, 84],
+            action_shape=6,
+            encoder_hidden_size_list=[128, 128, 256],
+            actor_head_hidden_size=256,
+            critic_head_hidden_size=256,
+        ),
+        learn=dict(
+            batch_size=128,
+            learning_rate=6e-4,
+            grad_clip_type='clip_norm',
+            clip_value=5,
+            value_weight=0.5,
+            entropy_weight=0.01,
+            discount_factor=0.99,
+            lambda_=0.95,
+            # hipGraph-capture the learn step (single-process only)
+            cuda_graph=os.environ.get('DING_IMPALA_GRAPH', '1') not in ('0', 'false'),
+        ),
     )
-    if graph_ok:
-        # hipGraph-captured fwd+v-trace+bwd (launch-bound at [33, 128])
-        if 'g' not in _IMPALA_GRAPH:
-            from ding.torch_utils.hip_graph import GraphedStep
-            _IMPALA_GRAPH['g'] = GraphedStep(step_fn)
-        _IMPALA_GRAPH['g']({'obs': obs, 'action': action, 'reward': reward})
-        optimizer.step()
-        return T * B
-    step_fn({'obs': obs, 'action': action, 'reward': reward})
-    if torch.distributed.is_initialized():
-        from ding.parallel import sync_gradients_flat
-        sync_gradients_flat(model)
-    optimizer.step()
+    cfg = EasyDict(deep_merge_dicts(cfg, user))
+    return _maybe_channels_last(IMPALAPolicy(cfg, enable_field=['learn']))
+
+
+def impala_step(policy, device: str, batch_size: int = 128, unroll_len: int = 32):
+    """One IMPALA learn iteration through IMPALAPolicy._forward_learn on a
+    synthetic device-resident time-major batch (the same-node trajectory
+    fast-path input format)."""
+    T, B = unroll_len, batch_size
+    obs_plus_1 = torch.rand(T + 1, B, 4, 84, 84, device=device)
+    if _channels_last_on():
+        obs_plus_1 = obs_plus_1.reshape((T + 1) * B, 4, 84, 84) \
+            .to(memory_format=torch.channels_last).view(T + 1, B, 4, 84, 84)
+    with torch.no_grad():
+        sample = policy._model(obs_plus_1[0], mode='compute_actor')['logit']
+    behaviour = sample.detach().unsqueeze(0).expand(T, B, -1).contiguous()
+    behaviour = behaviour + 0.1 * torch.randn_like(behaviour)
+    gumbel = -torch.log(-torch.log(torch.rand_like(behaviour) + 1e-10) + 1e-10)
+    action = (behaviour + gumbel).argmax(dim=-1)
+    batch = {
+        'obs_plus_1': obs_plus_1,
+        'logit': behaviour,
+        'action': action,
+        'reward': torch.randn(T, B, device=device),
+        'done': (torch.rand(T, B, device=device) < 0.002).float(),
+    }
+    policy._forward_learn(batch)
     return T * B
 
 
@@ -225,8 +227,8 @@ def main():
             'parallelism': f'dp{world_size}', 'minibatch': 320, 'epoch_per_collect': 10,
         }
     else:
-        model, optimizer = build_impala(device, multi_gpu=distributed)
-        step_fn = lambda: impala_step(model, optimizer, device)
+        policy = build_impala(device, multi_gpu=distributed)
+        step_fn = lambda: impala_step(policy, device)
         config = {
             'model': 'spaceinvaders_impala(conv[128,128,256] 4x84x84)', 'global_batch': 128 * 32 * world_size,
             'seq_len': 32, 'parallelism': f'dp{world_size}',

@@ -77,13 +77,21 @@ class IMPALAPolicy(Policy):
         self._graphed_step = None
         self._learn_model.reset()
 
-    def _data_preprocess_learn(self, data: List[Dict[str, Any]]):
-        """Collate unrolled samples -> time-major tensors [T, B, ...]."""
-        data = timestep_collate(data)
+    def _data_preprocess_learn(self, data):
+        """Collate unrolled samples -> time-major tensors [T, B, ...].
+
+        Accepts either a list of per-sample dicts (the collector path) or an
+        already-collated time-major dict whose tensors may live on the GPU
+        (the same-node trajectory fast path / bench path) — the latter skips
+        host collation entirely.
+        """
+        if not isinstance(data, dict):
+            data = timestep_collate(data)
         if self._cuda:
             data = to_device(data, self._device)
         data['weight'] = data.get('weight', None)
-        data['obs_plus_1'] = torch.cat([data['obs'], data['next_obs'][-1:]], dim=0).float()
+        if 'obs_plus_1' not in data:
+            data['obs_plus_1'] = torch.cat([data['obs'], data['next_obs'][-1:]], dim=0).float()
         return data
 
     def _learn_step(self, data: Dict[str, Any]) -> Dict[str, Any]:
