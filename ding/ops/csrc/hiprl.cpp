@@ -35,6 +35,15 @@ std::vector<torch::Tensor> ppo_bwd(
     torch::Tensor adv, torch::Tensor ret, torch::Tensor weight, torch::Tensor fwd_out, double clip_ratio,
     torch::Tensor grad_scales
 );
+std::vector<torch::Tensor> vtrace_fwd(
+    torch::Tensor t_logit, torch::Tensor b_logit, torch::Tensor action, torch::Tensor value,
+    torch::Tensor reward, torch::Tensor weight, double gamma, double lambda_, double rho_c, double c_c,
+    double rho_pg_c
+);
+std::vector<torch::Tensor> vtrace_bwd(
+    torch::Tensor t_logit, torch::Tensor row_out, torch::Tensor value, torch::Tensor ret, torch::Tensor adv,
+    torch::Tensor action, torch::Tensor weight, torch::Tensor gscales
+);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "DI-engine MI355X HIP kernels (gfx950)";
@@ -49,4 +58,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("stem_conv_wrw", &stem_conv_wrw, "direct 8x8s4 stem conv weight grad");
     m.def("lstm_cell_fwd", &lstm_cell_fwd, "fused LN-LSTM cell forward");
     m.def("lstm_cell_bwd", &lstm_cell_bwd, "fused LN-LSTM cell backward");
+    m.def("vtrace_fwd", &vtrace_fwd, "fused v-trace loss forward");
+    m.def("vtrace_bwd", &vtrace_bwd, "fused v-trace loss backward");
 }

@@ -145,6 +145,42 @@ def fused_ppo_error(logit_new, logit_old, action, value_new, value_old, adv, ret
     )
 
 
+class _FusedVtraceDiscrete(torch.autograd.Function):
+    """Fully fused discrete v-trace loss: 2 forward launches (row log-softmax/
+    IS/entropy + column scan with vs/adv/loss contribs), 1 analytic backward
+    launch. Parity: reference ding/rl_utils/vtrace.py:73."""
+
+    @staticmethod
+    def forward(ctx, t_logit, value, b_logit, action, reward, weight, gamma, lambda_, rho_c, c_c, rho_pg_c):
+        ext = _load()
+        w = weight if weight is not None else torch.empty(0, device=t_logit.device)
+        row_out, ret, adv, contrib = ext.vtrace_fwd(
+            t_logit.contiguous(), b_logit.contiguous(), action.contiguous().long(), value.contiguous(),
+            reward.contiguous(), w, float(gamma), float(lambda_), float(rho_c), float(c_c), float(rho_pg_c)
+        )
+        ctx.save_for_backward(t_logit, row_out, value, ret, adv, action, w)
+        losses = contrib.mean(dim=0)  # [3]: pg, value, entropy
+        return losses[0], losses[1], losses[2]
+
+    @staticmethod
+    def backward(ctx, g_pg, g_v, g_ent):
+        ext = _load()
+        t_logit, row_out, value, ret, adv, action, w = ctx.saved_tensors
+        # device-resident [3] so the launch is hipGraph-capture safe
+        gs = torch.stack([g_pg.reshape(()), g_v.reshape(()), g_ent.reshape(())]).float().contiguous()
+        d_logit, d_value = ext.vtrace_bwd(
+            t_logit.contiguous(), row_out, value.contiguous(), ret, adv, action.contiguous().long(), w, gs
+        )
+        return d_logit, d_value, None, None, None, None, None, None, None, None, None
+
+
+def fused_vtrace_error(t_logit, b_logit, action, value, reward, weight, gamma, lambda_, rho_c, c_c, rho_pg_c):
+    """Returns (policy_loss, value_loss, entropy_loss)."""
+    return _FusedVtraceDiscrete.apply(
+        t_logit, value, b_logit, action, reward, weight, gamma, lambda_, rho_c, c_c, rho_pg_c
+    )
+
+
 def scatter_connection(x, index, H: int, W: int, scatter_type: str):
     """x [B,M,N], index [B,M] flat spatial positions -> [B,N,H,W]."""
     ext = _load()

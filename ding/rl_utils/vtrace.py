@@ -56,6 +56,14 @@ def vtrace_error_discrete_action(
     value [T+1,B]; reward [T,B]. Returns (pg_loss, value_loss, entropy_loss).
     """
     target_output, behaviour_output, action, value, reward, weight = data
+    from ding.ops import dispatch
+    if dispatch.use_hip_autograd(target_output):
+        pg_loss, value_loss, entropy_loss = dispatch.fused_vtrace_error(
+            target_output.float(), behaviour_output.float(), action, value.float(), reward.float(),
+            None if weight is None else weight.float(), gamma, lambda_, rho_clip_ratio, c_clip_ratio,
+            rho_pg_clip_ratio
+        )
+        return vtrace_loss(pg_loss, value_loss, entropy_loss)
     with torch.no_grad():
         IS = compute_importance_weights(target_output, behaviour_output, action, 'discrete')
         rhos = IS.clamp(max=rho_clip_ratio)
