@@ -23,7 +23,9 @@ def default_collate(batch: Sequence, cat_1dim: bool = True, ignore_prefix: list 
     elem = batch[0]
     if isinstance(elem, torch.Tensor):
         if elem.shape == (1, ) and cat_1dim:
-            return torch.cat(batch, 0)
+            # tolerate mixed 0-dim / [1] entries (random-collect vs policy
+            # transitions can disagree on scalar action shape)
+            return torch.cat([b.reshape(1) for b in batch], 0)
         return torch.stack(batch, 0)
     if isinstance(elem, np.ndarray):
         return default_collate([torch.as_tensor(b) for b in batch], cat_1dim=cat_1dim)

@@ -1,0 +1,73 @@
+"""1-iteration smoke sweep over the dizoo config zoo (reference pattern:
+entry/tests/test_serial_entry.py:7-57). Every config must construct its env
++ policy and survive one collect->train iteration on CPU with shrunk sizes.
+"""
+import copy
+import importlib
+
+import pytest
+
+
+def _shrink(main_cfg, create_cfg):
+    """Scale a tuned config down to smoke size (tiny envs/batches, no eval)."""
+    m = copy.deepcopy(main_cfg)
+    c = copy.deepcopy(create_cfg)
+    m.exp_name = 'smoke_' + m.exp_name
+    m.env.collector_env_num = 2
+    m.env.evaluator_env_num = 1
+    m.env.n_evaluator_episode = 1
+    m.env.max_step = 30  # lite envs honour this cap
+    p = m.policy
+    p.cuda = False
+    if p.get('random_collect_size', 0) > 0:
+        p.random_collect_size = 16
+    if 'learn' in p:
+        if p.learn.get('update_per_collect') is not None:
+            p.learn.update_per_collect = 1
+        if p.learn.get('epoch_per_collect') is not None:
+            p.learn.epoch_per_collect = 1
+        if p.learn.get('batch_size') is not None:
+            p.learn.batch_size = min(p.learn.batch_size, 8)
+    if 'collect' in p and p.collect.get('n_sample') is not None:
+        p.collect.n_sample = max(int(p.get('unroll_len', 1)) * 2, 16)
+    if 'collect' in p and p.collect.get('n_episode') is not None:
+        p.collect.n_episode = 2
+    if 'other' in p and p.other.get('replay_buffer') is not None \
+            and p.other.replay_buffer.get('replay_buffer_size') is not None:
+        p.other.replay_buffer.replay_buffer_size = 1000
+    # keep subprocess manager machinery out of the smoke lane
+    c.env_manager.type = 'base'
+    return m, c
+
+
+# (config module, pipeline entry)
+SMOKE_CONFIGS = [
+    ('dizoo.box2d.lunarlander.config.lunarlander_dqn_config', 'serial'),
+    ('dizoo.box2d.lunarlander.config.lunarlander_ppo_config', 'onpolicy'),
+    ('dizoo.box2d.lunarlander.config.lunarlander_a2c_config', 'onpolicy'),
+    ('dizoo.box2d.lunarlander.config.lunarlander_impala_config', 'serial'),
+    ('dizoo.box2d.lunarlander.config.lunarlander_discrete_sac_config', 'serial'),
+    ('dizoo.box2d.lunarlander.config.lunarlander_cont_sac_config', 'serial'),
+    ('dizoo.box2d.lunarlander.config.lunarlander_cont_td3_config', 'serial'),
+    ('dizoo.box2d.bipedalwalker.config.bipedalwalker_sac_config', 'serial'),
+    ('dizoo.box2d.bipedalwalker.config.bipedalwalker_td3_config', 'serial'),
+]
+
+
+def _run_one(module_name: str, pipeline: str):
+    mod = importlib.import_module(module_name)
+    m, c = _shrink(mod.main_config, mod.create_config)
+    if pipeline == 'onpolicy':
+        from ding.entry import serial_pipeline_onpolicy
+        serial_pipeline_onpolicy((m, c), seed=0, max_train_iter=1)
+    elif pipeline == 'offline':
+        from ding.entry import serial_pipeline_offline
+        serial_pipeline_offline((m, c), seed=0, max_train_iter=1)
+    else:
+        from ding.entry import serial_pipeline
+        serial_pipeline((m, c), seed=0, max_train_iter=1)
+
+
+@pytest.mark.parametrize('module_name,pipeline', SMOKE_CONFIGS)
+def test_dizoo_config_smoke(module_name, pipeline):
+    _run_one(module_name, pipeline)
