@@ -51,6 +51,9 @@ SMOKE_CONFIGS = [
     ('dizoo.box2d.lunarlander.config.lunarlander_cont_td3_config', 'serial'),
     ('dizoo.box2d.bipedalwalker.config.bipedalwalker_sac_config', 'serial'),
     ('dizoo.box2d.bipedalwalker.config.bipedalwalker_td3_config', 'serial'),
+    ('dizoo.frozen_lake.config.frozen_lake_dqn_config', 'serial'),
+    ('dizoo.frozen_lake.config.frozen_lake_sql_config', 'serial'),
+    ('dizoo.taxi.config.taxi_dqn_config', 'serial'),
 ]
 
 
