@@ -54,6 +54,8 @@ SMOKE_CONFIGS = [
     ('dizoo.frozen_lake.config.frozen_lake_dqn_config', 'serial'),
     ('dizoo.frozen_lake.config.frozen_lake_sql_config', 'serial'),
     ('dizoo.taxi.config.taxi_dqn_config', 'serial'),
+    ('dizoo.pomdp.config.pomdp_dqn_config', 'serial'),
+    ('dizoo.pomdp.config.pomdp_ppo_config', 'serial'),
 ]
 
 
