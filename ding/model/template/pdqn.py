@@ -46,7 +46,11 @@ class PDQN(nn.Module):
             DiscreteHead(hid, self.action_type_shape, 1, activation=activation, norm_type=norm_type)
         )
 
-    def forward(self, inputs, mode: str) -> Dict:
+    def forward(self, inputs, mode: str = None) -> Dict:
+        if mode is None:
+            # whole-model pass for collect/eval wrappers: args net then Q net
+            args = self.compute_continuous(inputs)['action_args']
+            return self.compute_discrete({'state': inputs, 'action_args': args})
         assert mode in self.mode
         return getattr(self, mode)(inputs)
 

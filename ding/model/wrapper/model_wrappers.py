@@ -238,8 +238,13 @@ class HybridArgmaxSampleWrapper(IModelWrapper):
     def forward(self, *args, **kwargs):
         output = self._model.forward(*args, **kwargs)
         logit = output['logit']
-        action_type = logit['action_type'].argmax(dim=-1)
-        action_args = logit['action_args']['mu'] if isinstance(logit['action_args'], dict) else output['action_args']
+        if isinstance(logit, dict):  # VAC-style {'action_type','action_args'}
+            action_type = logit['action_type'].argmax(dim=-1)
+            action_args = logit['action_args']['mu'] if isinstance(logit['action_args'], dict) \
+                else output['action_args']
+        else:  # PDQN-style flat Q logits + separate args
+            action_type = logit.argmax(dim=-1)
+            action_args = output['action_args']
         output['action'] = {'action_type': action_type, 'action_args': action_args}
         return output
 
@@ -247,7 +252,7 @@ class HybridArgmaxSampleWrapper(IModelWrapper):
 class HybridEpsGreedySampleWrapper(IModelWrapper):
 
     def forward(self, *args, **kwargs):
-        eps = kwargs.pop('eps')
+        eps = kwargs.pop('eps', 0.0)
         output = self._model.forward(*args, **kwargs)
         logit = output['logit']['action_type'] if isinstance(output['logit'], dict) else output['logit']
         greedy = logit.argmax(dim=-1)

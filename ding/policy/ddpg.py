@@ -93,7 +93,13 @@ class DDPGPolicy(Policy):
         q_value = self._learn_model.forward({'obs': data['obs'], 'action': data['action']}, mode='compute_critic')['q_value']
         with torch.no_grad():
             next_actor_out = self._target_model.forward(data['next_obs'], mode='compute_actor')
-            next_action = next_actor_out['action']
+            if 'action' in next_actor_out:
+                next_action = next_actor_out['action']
+            else:  # hybrid actor: compose greedy type + args
+                next_action = {
+                    'action_type': next_actor_out['logit'].argmax(-1),
+                    'action_args': next_actor_out['action_args'],
+                }
             next_data = {'obs': data['next_obs'], 'action': next_action}
             target_q = self._target_model.forward(next_data, mode='compute_critic')['q_value']
         reward = data['reward']
@@ -122,7 +128,14 @@ class DDPGPolicy(Policy):
         actor_loss = torch.zeros(())
         if self._forward_learn_cnt % self._actor_update_freq == 0:
             actor_out = self._learn_model.forward(data['obs'], mode='compute_actor')
-            actor_data = {'obs': data['obs'], 'action': actor_out['action']}
+            if 'action' in actor_out:
+                actor_data = {'obs': data['obs'], 'action': actor_out['action']}
+            else:  # hybrid: soft type probs would break the critic cat; use logit passthrough
+                actor_data = {
+                    'obs': data['obs'], 'logit': actor_out['logit'],
+                    'action': {'action_type': actor_out['logit'].argmax(-1),
+                               'action_args': actor_out['action_args']},
+                }
             q = self._learn_model.forward(actor_data, mode='compute_critic')['q_value']
             if self._twin_critic:
                 q = q[0]
@@ -149,10 +162,13 @@ class DDPGPolicy(Policy):
 
     def _init_collect(self) -> None:
         self._unroll_len = self._cfg.collect.unroll_len
-        self._collect_model = model_wrap(
-            self._model, wrapper_name='action_noise', noise_type='gauss',
-            noise_kwargs={'mu': 0.0, 'sigma': self._cfg.collect.noise_sigma}, noise_range=None
-        )
+        if self._cfg.get('action_space', None) == 'hybrid':  # PADDPG lane
+            self._collect_model = model_wrap(self._model, wrapper_name='hybrid_eps_greedy_multinomial_sample')
+        else:
+            self._collect_model = model_wrap(
+                self._model, wrapper_name='action_noise', noise_type='gauss',
+                noise_kwargs={'mu': 0.0, 'sigma': self._cfg.collect.noise_sigma}, noise_range=None
+            )
         self._collect_model.reset()
 
     def _forward_collect(self, data: Dict[int, Any], **kwargs) -> Dict[int, Any]:
@@ -181,7 +197,10 @@ class DDPGPolicy(Policy):
         return get_train_sample(transitions, self._unroll_len)
 
     def _init_eval(self) -> None:
-        self._eval_model = model_wrap(self._model, wrapper_name='base')
+        if self._cfg.get('action_space', None) == 'hybrid':
+            self._eval_model = model_wrap(self._model, wrapper_name='hybrid_argmax_sample')
+        else:
+            self._eval_model = model_wrap(self._model, wrapper_name='base')
         self._eval_model.reset()
 
     def _forward_eval(self, data: Dict[int, Any]) -> Dict[int, Any]:

@@ -56,6 +56,10 @@ SMOKE_CONFIGS = [
     ('dizoo.taxi.config.taxi_dqn_config', 'serial'),
     ('dizoo.pomdp.config.pomdp_dqn_config', 'serial'),
     ('dizoo.pomdp.config.pomdp_ppo_config', 'serial'),
+    ('dizoo.gym_hybrid.config.gym_hybrid_pdqn_config', 'serial'),
+    ('dizoo.gym_hybrid.config.gym_hybrid_mpdqn_config', 'serial'),
+    ('dizoo.gym_hybrid.config.gym_hybrid_hppo_config', 'onpolicy'),
+    ('dizoo.gym_hybrid.config.gym_hybrid_ddpg_config', 'serial'),
 ]
 
 
