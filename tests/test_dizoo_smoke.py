@@ -60,6 +60,15 @@ SMOKE_CONFIGS = [
     ('dizoo.gym_hybrid.config.gym_hybrid_mpdqn_config', 'serial'),
     ('dizoo.gym_hybrid.config.gym_hybrid_hppo_config', 'onpolicy'),
     ('dizoo.gym_hybrid.config.gym_hybrid_ddpg_config', 'serial'),
+    ('dizoo.petting_zoo.config.ptz_simple_spread_qmix_config', 'serial'),
+    ('dizoo.petting_zoo.config.ptz_simple_spread_vdn_config', 'serial'),
+    ('dizoo.petting_zoo.config.ptz_simple_spread_wqmix_config', 'serial'),
+    ('dizoo.petting_zoo.config.ptz_simple_spread_qtran_config', 'serial'),
+    ('dizoo.petting_zoo.config.ptz_simple_spread_collaq_config', 'serial'),
+    ('dizoo.petting_zoo.config.ptz_simple_spread_coma_config', 'serial'),
+    ('dizoo.petting_zoo.config.ptz_simple_spread_mappo_config', 'onpolicy'),
+    ('dizoo.petting_zoo.config.ptz_simple_spread_happo_config', 'onpolicy'),
+    ('dizoo.petting_zoo.config.ptz_simple_spread_madqn_config', 'serial'),
 ]
 
 
