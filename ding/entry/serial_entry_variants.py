@@ -30,7 +30,9 @@ def _build_workers(cfg, policy, collector_env, evaluator_env):
     evaluator = InteractionSerialEvaluator(
         cfg.policy.eval.evaluator, evaluator_env, policy.eval_mode, exp_name=cfg.exp_name
     )
-    replay_buffer = create_buffer(cfg.policy.other.replay_buffer, exp_name=cfg.exp_name)
+    rb_cfg = cfg.policy.get('other', EasyDict({})).get('replay_buffer', None)
+    # on-policy cfgs have no buffer section; variants that need one get None
+    replay_buffer = create_buffer(rb_cfg, exp_name=cfg.exp_name) if rb_cfg is not None else None
     commander = BaseSerialCommander(EasyDict({}), learner, collector, evaluator, replay_buffer, policy.command_mode)
     return learner, collector, evaluator, replay_buffer, commander
 

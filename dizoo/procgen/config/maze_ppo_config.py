@@ -1,0 +1,45 @@
+"""Procgen maze on-policy PPO (reference maze_ppo_config.py)."""
+from ding.utils import EasyDict
+
+maze_ppo_config = EasyDict(dict(
+    exp_name='maze_ppo_seed0',
+    env=dict(
+        env_id='maze',
+        collector_env_num=16,
+        evaluator_env_num=4,
+        n_evaluator_episode=4,
+        stop_value=10,
+    ),
+    policy=dict(
+        cuda=True,
+        action_space='discrete',
+        recompute_adv=True,
+        model=dict(
+            obs_shape=[3, 64, 64],
+            action_shape=15,
+            action_space='discrete',
+            encoder_hidden_size_list=[32, 64, 64, 128],
+            actor_head_hidden_size=128,
+            critic_head_hidden_size=128,
+        ),
+        learn=dict(
+            epoch_per_collect=3,
+            batch_size=512,
+            learning_rate=5e-4,
+            value_weight=0.5,
+            entropy_weight=0.01,
+            clip_ratio=0.2,
+            adv_norm=True,
+            value_norm=True,
+        ),
+        collect=dict(n_sample=4096, unroll_len=1, discount_factor=0.999, gae_lambda=0.95),
+        eval=dict(evaluator=dict(eval_freq=5000, )),
+    ),
+))
+main_config = maze_ppo_config
+maze_ppo_create_config = EasyDict(dict(
+    env=dict(type='procgen', import_names=['dizoo.procgen.envs.procgen_env']),
+    env_manager=dict(type='subprocess'),
+    policy=dict(type='ppo'),
+))
+create_config = maze_ppo_create_config

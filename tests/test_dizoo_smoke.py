@@ -69,6 +69,13 @@ SMOKE_CONFIGS = [
     ('dizoo.petting_zoo.config.ptz_simple_spread_mappo_config', 'onpolicy'),
     ('dizoo.petting_zoo.config.ptz_simple_spread_happo_config', 'onpolicy'),
     ('dizoo.petting_zoo.config.ptz_simple_spread_madqn_config', 'serial'),
+    ('dizoo.procgen.config.coinrun_dqn_config', 'serial'),
+    ('dizoo.procgen.config.coinrun_ppo_config', 'onpolicy'),
+    ('dizoo.procgen.config.coinrun_ppg_config', 'onpolicy_ppg'),
+    ('dizoo.procgen.config.maze_dqn_config', 'serial'),
+    ('dizoo.procgen.config.maze_ppo_config', 'onpolicy'),
+    ('dizoo.procgen.config.bigfish_ppg_config', 'onpolicy_ppg'),
+    ('dizoo.procgen.config.bigfish_plr_config', 'plr'),
 ]
 
 
@@ -78,6 +85,12 @@ def _run_one(module_name: str, pipeline: str):
     if pipeline == 'onpolicy':
         from ding.entry import serial_pipeline_onpolicy
         serial_pipeline_onpolicy((m, c), seed=0, max_train_iter=1)
+    elif pipeline == 'onpolicy_ppg':
+        from ding.entry import serial_pipeline_onpolicy_ppg
+        serial_pipeline_onpolicy_ppg((m, c), seed=0, max_train_iter=1)
+    elif pipeline == 'plr':
+        from ding.entry import serial_pipeline_plr
+        serial_pipeline_plr((m, c), seed=0, max_train_iter=1)
     elif pipeline == 'offline':
         from ding.entry import serial_pipeline_offline
         serial_pipeline_offline((m, c), seed=0, max_train_iter=1)
