@@ -76,12 +76,48 @@ SMOKE_CONFIGS = [
     ('dizoo.procgen.config.maze_ppo_config', 'onpolicy'),
     ('dizoo.procgen.config.bigfish_ppg_config', 'onpolicy_ppg'),
     ('dizoo.procgen.config.bigfish_plr_config', 'plr'),
+    ('dizoo.d4rl.config.hopper_medium_td3bc_config', 'offline'),
+    ('dizoo.d4rl.config.hopper_medium_cql_config', 'offline'),
+    ('dizoo.d4rl.config.halfcheetah_medium_iql_config', 'offline'),
+    ('dizoo.d4rl.config.walker2d_medium_edac_config', 'offline'),
+    ('dizoo.d4rl.config.hopper_medium_bcq_config', 'offline'),
+    ('dizoo.d4rl.config.hopper_expert_dt_config', 'offline_dt'),
 ]
 
 
-def _run_one(module_name: str, pipeline: str):
+def _run_one(module_name: str, pipeline: str, tmp_dir: str = None):
     mod = importlib.import_module(module_name)
     m, c = _shrink(mod.main_config, mod.create_config)
+    if module_name.startswith('dizoo.d4rl'):
+        import os
+        from dizoo.d4rl.generate import generate_d4rl_npz
+        path = os.path.join(tmp_dir or '.', m.env.env_id + '.npz')
+        generate_d4rl_npz(m.env.env_id, path, n_transitions=512)
+        if 'dataset' in m:
+            m.dataset.data_dir_prefix = path
+        if m.policy.get('collect', {}).get('data_path'):
+            m.policy.collect.data_path = path
+    if pipeline == 'offline_dt':
+        # DT trains through the task-pipeline trainer on the trajectory dataset
+        import torch
+        from ding.utils.data.dataset import D4RLTrajectoryDataset
+        from ding.framework import OfflineRLContext, task
+        from ding.framework.middleware import offline_data_fetcher, trainer
+        from ding.policy import create_policy
+        from ding.policy.dt import DTPolicy
+        from ding.utils import EasyDict, deep_merge_dicts
+        pcfg = EasyDict(deep_merge_dicts(DTPolicy.default_config(), m.policy))
+        pcfg.type = c.policy.type
+        pcfg.cuda = False
+        pcfg.learn.batch_size = 8
+        m.policy = pcfg
+        dataset = D4RLTrajectoryDataset(m)
+        policy = create_policy(pcfg, enable_field=['learn'])
+        with task.start(ctx=OfflineRLContext()):
+            task.use(offline_data_fetcher(m, dataset))
+            task.use(trainer(m, policy.learn_mode))
+            task.run(max_step=2)
+        return
     if pipeline == 'onpolicy':
         from ding.entry import serial_pipeline_onpolicy
         serial_pipeline_onpolicy((m, c), seed=0, max_train_iter=1)
@@ -100,5 +136,5 @@ def _run_one(module_name: str, pipeline: str):
 
 
 @pytest.mark.parametrize('module_name,pipeline', SMOKE_CONFIGS)
-def test_dizoo_config_smoke(module_name, pipeline):
-    _run_one(module_name, pipeline)
+def test_dizoo_config_smoke(module_name, pipeline, tmp_path):
+    _run_one(module_name, pipeline, tmp_dir=str(tmp_path))
