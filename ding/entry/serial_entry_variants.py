@@ -66,15 +66,22 @@ def serial_pipeline_reward_model(
         if cooptrain_reward:
             reward_model.collect_data(new_data)
             reward_model.train()
-        replay_buffer.push(new_data, cur_collector_envstep=collector.envstep)
-        for i in range(cfg.policy.learn.update_per_collect):
-            train_data = replay_buffer.sample(learner.policy.get_attribute('batch_size'), learner.train_iter)
-            if train_data is None:
-                break
-            train_data = reward_model.estimate(train_data)
+        if cfg.policy.on_policy:
+            # on-policy lane (e.g. RND + onpolicy PPO): relabel and train on
+            # the whole fresh batch, no buffer (reference serial_entry_onpolicy
+            # + reward-model estimate)
+            train_data = reward_model.estimate(new_data)
             learner.train(train_data, collector.envstep)
-            if learner.policy.get_attribute('priority'):
-                replay_buffer.update(learner.priority_info)
+        else:
+            replay_buffer.push(new_data, cur_collector_envstep=collector.envstep)
+            for i in range(cfg.policy.learn.update_per_collect):
+                train_data = replay_buffer.sample(learner.policy.get_attribute('batch_size'), learner.train_iter)
+                if train_data is None:
+                    break
+                train_data = reward_model.estimate(train_data)
+                learner.train(train_data, collector.envstep)
+                if learner.policy.get_attribute('priority'):
+                    replay_buffer.update(learner.priority_info)
         if cooptrain_reward:
             reward_model.clear_data()
         if collector.envstep >= max_env_step or learner.train_iter >= max_train_iter:

@@ -65,7 +65,12 @@ def serial_pipeline_ngu(
             nu = cfg.policy.get('nu', 0.01)
             for d, r_i in zip(train_data, intrinsic):
                 r = d['reward']
-                d['reward'] = r + nu * r_i.to(r.dtype) if isinstance(r, torch.Tensor) else float(r) + nu * float(r_i)
+                if isinstance(r, (list, tuple)):  # r2d2 unroll: spread over steps
+                    d['reward'] = [torch.as_tensor(x, dtype=torch.float32) + nu * float(r_i) for x in r]
+                elif isinstance(r, torch.Tensor):
+                    d['reward'] = r + nu * r_i.to(r.dtype)
+                else:
+                    d['reward'] = float(r) + nu * float(r_i)
             learner.train(train_data, collector.envstep)
             if learner.policy.get_attribute('priority'):
                 replay_buffer.update(learner.priority_info)

@@ -42,7 +42,14 @@ class RndNGURewardModel(BaseRewardModel):
         self._rms = RunningMeanStd(epsilon=1e-4)
 
     def collect_data(self, data: list) -> None:
-        self.train_obs.extend([d['obs'] for d in data])
+        for d in data:
+            obs = d['obs']
+            if isinstance(obs, (list, tuple)):  # r2d2 unroll sample: [T] of obs
+                self.train_obs.extend(obs)
+            elif isinstance(obs, torch.Tensor) and obs.dim() == 2:  # [T, obs]
+                self.train_obs.extend(obs.unbind(0))
+            else:
+                self.train_obs.append(obs)
 
     def clear_data(self) -> None:
         self.train_obs = []
@@ -59,9 +66,19 @@ class RndNGURewardModel(BaseRewardModel):
             loss.backward()
             self.opt.step()
 
+    @staticmethod
+    def _sample_obs(d):
+        """One representative obs per sample: sequences (r2d2 unrolls) use the
+        mean over the window."""
+        obs = d['obs']
+        if isinstance(obs, (list, tuple)):
+            return torch.stack([torch.as_tensor(o, dtype=torch.float32) for o in obs]).mean(0)
+        obs = torch.as_tensor(obs, dtype=torch.float32)
+        return obs.mean(0) if obs.dim() == 2 else obs
+
     def estimate(self, data: list) -> torch.Tensor:
         """Return the alpha multiplier per transition (not reward rewrite)."""
-        obs = torch.stack([torch.as_tensor(d['obs'], dtype=torch.float32) for d in data]).to(self.device)
+        obs = torch.stack([self._sample_obs(d) for d in data]).to(self.device)
         with torch.no_grad():
             p, t = self.reward_model(obs)
             err = (p - t).pow(2).sum(1)
@@ -100,7 +117,22 @@ class EpisodicNGURewardModel(BaseRewardModel):
         self._episode_memory = defaultdict(list)  # env_id -> embeddings
 
     def collect_data(self, data: list) -> None:
-        self.train_data.extend(data)
+        for d in data:
+            obs = d['obs']
+            if isinstance(obs, (list, tuple)) or (isinstance(obs, torch.Tensor) and obs.dim() == 2):
+                # r2d2 unroll sample: explode the [T] sequence into transitions
+                T = len(obs)
+                next_obs = d.get('next_obs')  # r2d2 unrolls have none: shift obs
+                for t in range(T):
+                    nxt = next_obs[t] if next_obs is not None else obs[min(t + 1, T - 1)]
+                    self.train_data.append({
+                        'obs': obs[t],
+                        'next_obs': nxt,
+                        'action': d['action'][t],
+                        'env_id': d.get('env_id', 0),
+                    })
+            else:
+                self.train_data.append(d)
 
     def clear_data(self) -> None:
         self.train_data = []
@@ -129,8 +161,11 @@ class EpisodicNGURewardModel(BaseRewardModel):
         rewards = []
         with torch.no_grad():
             for d in data:
-                env_id = int(d.get('env_id', 0))
-                obs = torch.as_tensor(d['obs'], dtype=torch.float32).unsqueeze(0).to(self.device)
+                env_id = d.get('env_id', 0)
+                if isinstance(env_id, torch.Tensor):
+                    env_id = int(env_id.reshape(-1)[0])
+                env_id = int(env_id)
+                obs = RndNGURewardModel._sample_obs(d).unsqueeze(0).to(self.device)
                 e = self.embed(obs).squeeze(0)
                 mem = self._episode_memory[env_id]
                 if len(mem) == 0:

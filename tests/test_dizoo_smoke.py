@@ -82,6 +82,11 @@ SMOKE_CONFIGS = [
     ('dizoo.d4rl.config.walker2d_medium_edac_config', 'offline'),
     ('dizoo.d4rl.config.hopper_medium_bcq_config', 'offline'),
     ('dizoo.d4rl.config.hopper_expert_dt_config', 'offline_dt'),
+    ('dizoo.minigrid.config.minigrid_onppo_config', 'onpolicy'),
+    ('dizoo.minigrid.config.minigrid_r2d2_config', 'serial'),
+    ('dizoo.minigrid.config.minigrid_rnd_onppo_config', 'reward_model'),
+    ('dizoo.minigrid.config.minigrid_icm_offppo_config', 'reward_model'),
+    ('dizoo.minigrid.config.minigrid_ngu_config', 'ngu'),
 ]
 
 
@@ -127,6 +132,12 @@ def _run_one(module_name: str, pipeline: str, tmp_dir: str = None):
     elif pipeline == 'plr':
         from ding.entry import serial_pipeline_plr
         serial_pipeline_plr((m, c), seed=0, max_train_iter=1)
+    elif pipeline == 'reward_model':
+        from ding.entry import serial_pipeline_reward_model
+        serial_pipeline_reward_model((m, c), seed=0, max_train_iter=1)
+    elif pipeline == 'ngu':
+        from ding.entry import serial_pipeline_ngu
+        serial_pipeline_ngu((m, c), seed=0, max_train_iter=1)
     elif pipeline == 'offline':
         from ding.entry import serial_pipeline_offline
         serial_pipeline_offline((m, c), seed=0, max_train_iter=1)
