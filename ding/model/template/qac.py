@@ -123,7 +123,7 @@ class ContinuousQAC(nn.Module):
             self.critic = self.critic_head
 
     def _enc(self, x, role: str):
-        if self.encoder is None and not hasattr(self, 'actor_encoder'):
+        if getattr(self, 'encoder', None) is None and not hasattr(self, 'actor_encoder'):
             return x
         if self.share_encoder:
             return self.encoder(x)

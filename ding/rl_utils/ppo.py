@@ -172,7 +172,9 @@ def ppo_error_continuous(
     use_value_clip: bool = True,
     dual_clip: Optional[float] = None,
 ) -> Tuple[namedtuple, namedtuple]:
-    mu_sigma_new, mu_sigma_old, action, value_new, value_old, adv, return_, weight = data
+    # accept both ppo_data (9 fields, trailing logit_pretrained) and
+    # ppo_data_continuous (8 fields)
+    mu_sigma_new, mu_sigma_old, action, value_new, value_old, adv, return_, weight = tuple(data)[:8]
     pol, info = ppo_policy_error_continuous(
         ppo_policy_data_continuous(mu_sigma_new, mu_sigma_old, action, adv, weight), clip_ratio, dual_clip
     )

@@ -87,6 +87,10 @@ SMOKE_CONFIGS = [
     ('dizoo.minigrid.config.minigrid_rnd_onppo_config', 'reward_model'),
     ('dizoo.minigrid.config.minigrid_icm_offppo_config', 'reward_model'),
     ('dizoo.minigrid.config.minigrid_ngu_config', 'ngu'),
+    ('dizoo.slime_volley.config.slime_volley_ppo_config', 'onpolicy'),
+    ('dizoo.dmc2gym.config.dmc2gym_sac_state_config', 'serial'),
+    ('dizoo.dmc2gym.config.dmc2gym_sac_pixel_config', 'serial'),
+    ('dizoo.dmc2gym.config.dmc2gym_ppo_config', 'onpolicy'),
 ]
 
 
@@ -149,3 +153,39 @@ def _run_one(module_name: str, pipeline: str, tmp_dir: str = None):
 @pytest.mark.parametrize('module_name,pipeline', SMOKE_CONFIGS)
 def test_dizoo_config_smoke(module_name, pipeline, tmp_path):
     _run_one(module_name, pipeline, tmp_dir=str(tmp_path))
+
+
+def test_slime_volley_agent_vs_agent_battle():
+    """agent_vs_agent mode feeds the 1v1 battle collector (league lane)."""
+    import numpy as np
+    from dizoo.slime_volley.envs.slime_volley_env import SlimeVolleyEnv
+    env = SlimeVolleyEnv({'agent_vs_agent': True, 'max_step': 200})
+    env.seed(0)
+    obs = env.reset()
+    assert isinstance(obs, list) and obs[0].shape == (12, ) and obs[1].shape == (12, )
+    for _ in range(200):
+        ts = env.step([env.random_action(), env.random_action()])
+        assert ts.reward.shape == (2, )
+        assert abs(float(ts.reward[0]) + float(ts.reward[1])) < 1e-6  # zero-sum
+        if ts.done:
+            assert ts.info['result'] in ('wins', 'losses', 'draws')
+            break
+    assert ts.done
+
+
+def test_dmc2gym_pixel_render_tracks_state():
+    import numpy as np
+    from dizoo.dmc2gym.envs.dmc2gym_env import DMC2GymEnv
+    env = DMC2GymEnv({'domain_name': 'cartpole', 'task_name': 'swingup', 'from_pixels': True, 'frame_skip': 2})
+    env.seed(0)
+    obs = env.reset()
+    assert obs.shape == (3, 84, 84) and obs.dtype == np.float32
+    ts = env.step(np.array([1.0]))
+    assert ts.obs.shape == (3, 84, 84)
+    assert 0.0 <= float(ts.reward[0]) <= 1.0
+    # swingup starts hanging: reward must be near zero, upright near one
+    env2 = DMC2GymEnv({'domain_name': 'cartpole', 'task_name': 'balance'})
+    env2.seed(0)
+    env2.reset()
+    r_bal = float(env2.step(np.array([0.0])).reward[0])
+    assert r_bal > 0.5
