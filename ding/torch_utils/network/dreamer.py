@@ -86,12 +86,12 @@ class OneHotDist(torchd.one_hot_categorical.OneHotCategorical):
 
     def __init__(self, logits=None, probs=None, unimix_ratio: float = 0.0):
         if logits is not None and unimix_ratio > 0.0:
-            probs = F.softmax(logits, dim=-1)
-            probs = probs * (1.0 - unimix_ratio) + unimix_ratio / probs.shape[-1]
-            logits = torch.log(probs)
-            super().__init__(logits=logits, probs=None)
-        else:
-            super().__init__(logits=logits, probs=probs)
+            # blend the categorical with a uniform floor:
+            # p <- (1-u) softmax(z) + u/K, re-expressed as logits
+            u = unimix_ratio
+            mixed = (1.0 - u) * F.softmax(logits, dim=-1) + u / logits.shape[-1]
+            logits, probs = mixed.log(), None
+        super().__init__(logits=logits, probs=probs)
 
     def mode(self):
         _mode = F.one_hot(torch.argmax(super().logits, axis=-1), super().logits.shape[-1])
@@ -124,24 +124,22 @@ class TwoHotDistSymlog:
     def mode(self):
         return self.mean()
 
+    def _twohot(self, x: torch.Tensor) -> torch.Tensor:
+        """Two-hot encoding of symlog values via bucketize: mass split between
+        the two neighbouring bins, inversely proportional to distance."""
+        K = self.buckets.numel()
+        hi = torch.bucketize(x, self.buckets).clamp_(0, K - 1)
+        lo = (hi - 1).clamp_(0, K - 1)
+        d_lo = (x - self.buckets[lo]).abs()
+        d_hi = (self.buckets[hi] - x).abs()
+        span = (d_lo + d_hi).clamp_min(1e-8)
+        # out-of-range x collapses to a single bin (lo == hi -> both halves
+        # land on the same one-hot and sum to 1)
+        return F.one_hot(lo, K) * (d_hi / span)[..., None] + F.one_hot(hi, K) * (d_lo / span)[..., None]
+
     def log_prob(self, x: torch.Tensor) -> torch.Tensor:
-        x = symlog(x)
-        below = (self.buckets <= x[..., None]).to(torch.int32).sum(dim=-1) - 1
-        above = len(self.buckets) - (self.buckets > x[..., None]).to(torch.int32).sum(dim=-1)
-        below = torch.clip(below, 0, len(self.buckets) - 1)
-        above = torch.clip(above, 0, len(self.buckets) - 1)
-        equal = below == above
-        dist_to_below = torch.where(equal, 1, torch.abs(self.buckets[below] - x))
-        dist_to_above = torch.where(equal, 1, torch.abs(self.buckets[above] - x))
-        total = dist_to_below + dist_to_above
-        weight_below = dist_to_above / total
-        weight_above = dist_to_below / total
-        target = (
-            F.one_hot(below, num_classes=len(self.buckets)) * weight_below[..., None] +
-            F.one_hot(above, num_classes=len(self.buckets)) * weight_above[..., None]
-        )
+        target = self._twohot(symlog(x)).squeeze(-2)
         log_pred = self.logits - torch.logsumexp(self.logits, -1, keepdim=True)
-        target = target.squeeze(-2)
         return (target * log_pred).sum(-1)
 
     def log_prob_target(self, target: torch.Tensor) -> torch.Tensor:
@@ -415,40 +413,29 @@ class ActionHead(nn.Module):
 
 
 def static_scan(fn: Callable, inputs: Tuple, start, reverse: bool = False):
-    """Unrolled scan: apply `fn(carry, *inputs_t)` along dim 0, stacking every
-    output. `start` may be a (nested) tuple of dict/tensor carries."""
-    last = start
-    flag = True
-    indices = range(inputs[0].shape[0])
-    outputs = None
-    for index in indices:
-        inp = (lambda x: (_inp[x] for _inp in inputs))(index)
-        last = fn(last, *inp)
-        if flag:
-            if isinstance(last, dict):
-                outputs = {key: value.clone().unsqueeze(0) for key, value in last.items()}
-            else:
-                outputs = []
-                for _last in last:
-                    if isinstance(_last, dict):
-                        outputs.append({key: value.clone().unsqueeze(0) for key, value in _last.items()})
-                    else:
-                        outputs.append(_last.clone().unsqueeze(0))
-            flag = False
-        else:
-            if isinstance(last, dict):
-                for key in last.keys():
-                    outputs[key] = torch.cat([outputs[key], last[key].unsqueeze(0)], dim=0)
-            else:
-                for j in range(len(outputs)):
-                    if isinstance(last[j], dict):
-                        for key in last[j].keys():
-                            outputs[j][key] = torch.cat([outputs[j][key], last[j][key].unsqueeze(0)], dim=0)
-                    else:
-                        outputs[j] = torch.cat([outputs[j], last[j].unsqueeze(0)], dim=0)
-    if isinstance(outputs, dict):
-        outputs = [outputs]
-    return outputs
+    """Unrolled scan: apply ``fn(carry, *inputs_t)`` along dim 0 and stack the
+    per-step outputs once at the end (one torch.stack per leaf instead of a
+    cat per step — O(T) not O(T^2) copies).
+
+    The carry may be a tensor, a dict of tensors, or a tuple mixing both;
+    always returns a list of stacked structures (a lone dict carry comes back
+    as ``[dict]``).
+    """
+    steps = []
+    carry = start
+    for t in range(inputs[0].shape[0]):
+        carry = fn(carry, *(seq[t] for seq in inputs))
+        steps.append(carry)
+
+    def stack_leaves(get):
+        sample = get(steps[0])
+        if isinstance(sample, dict):
+            return {k: torch.stack([get(s)[k] for s in steps], dim=0) for k in sample}
+        return torch.stack([get(s) for s in steps], dim=0)
+
+    if isinstance(steps[0], dict):
+        return [stack_leaves(lambda s: s)]
+    return [stack_leaves(lambda s, j=j: s[j]) for j in range(len(steps[0]))]
 
 
 class GRUCellLN(nn.Module):
@@ -480,48 +467,49 @@ class GRUCellLN(nn.Module):
         return output, [output]
 
 
+# trunc_normal_(-2sigma, 2sigma) keeps ~87.96% of the mass; dividing the std by
+# this factor restores unit variance after truncation
+_TRUNC_STD_CORRECTION = 0.87962566103423978
+
+
+def _fan_avg(m: nn.Module) -> float:
+    """(fan_in + fan_out) / 2, counting the kernel area for convolutions."""
+    if isinstance(m, (nn.Conv2d, nn.ConvTranspose2d)):
+        area = m.kernel_size[0] * m.kernel_size[1]
+        return area * (m.in_channels + m.out_channels) / 2.0
+    return (m.in_features + m.out_features) / 2.0
+
+
+def _zero_bias(m: nn.Module) -> None:
+    if getattr(m, 'bias', None) is not None:
+        nn.init.zeros_(m.bias)
+
+
 def weight_init(m):
-    """He-style init with layer-norm reset (DreamerV3 default)."""
-    if isinstance(m, nn.Linear):
-        in_num = m.in_features
-        out_num = m.out_features
-        denoms = (in_num + out_num) / 2.0
-        scale = 1.0 / denoms
-        std = np.sqrt(scale) / 0.87962566103423978
-        nn.init.trunc_normal_(m.weight.data, mean=0.0, std=std, a=-2.0 * std, b=2.0 * std)
-        if hasattr(m.bias, 'data'):
-            m.bias.data.fill_(0.0)
-    elif isinstance(m, (nn.Conv2d, nn.ConvTranspose2d)):
-        space = m.kernel_size[0] * m.kernel_size[1]
-        in_num = space * m.in_channels
-        out_num = space * m.out_channels
-        denoms = (in_num + out_num) / 2.0
-        scale = 1.0 / denoms
-        std = np.sqrt(scale) / 0.87962566103423978
-        nn.init.trunc_normal_(m.weight.data, mean=0.0, std=std, a=-2.0, b=2.0)
-        if hasattr(m.bias, 'data'):
-            m.bias.data.fill_(0.0)
-    elif isinstance(m, nn.LayerNorm):
-        m.weight.data.fill_(1.0)
-        if hasattr(m.bias, 'data'):
-            m.bias.data.fill_(0.0)
+    """Variance-scaling truncated-normal init (DreamerV3 default), with
+    layer-norms reset to identity."""
+    if isinstance(m, nn.LayerNorm):
+        nn.init.ones_(m.weight)
+        _zero_bias(m)
+        return
+    if not isinstance(m, (nn.Linear, nn.Conv2d, nn.ConvTranspose2d)):
+        return
+    std = math.sqrt(1.0 / _fan_avg(m)) / _TRUNC_STD_CORRECTION
+    bound = 2.0 * std if isinstance(m, nn.Linear) else 2.0
+    nn.init.trunc_normal_(m.weight, std=std, a=-bound, b=bound)
+    _zero_bias(m)
 
 
 def uniform_weight_init(given_scale: float):
+    """Variance-scaling uniform init (used for output heads, often scale 0)."""
 
     def f(m):
-        if isinstance(m, nn.Linear):
-            in_num = m.in_features
-            out_num = m.out_features
-            denoms = (in_num + out_num) / 2.0
-            scale = given_scale / denoms
-            limit = np.sqrt(3 * scale)
-            nn.init.uniform_(m.weight.data, a=-limit, b=limit)
-            if hasattr(m.bias, 'data'):
-                m.bias.data.fill_(0.0)
-        elif isinstance(m, nn.LayerNorm):
-            m.weight.data.fill_(1.0)
-            if hasattr(m.bias, 'data'):
-                m.bias.data.fill_(0.0)
+        if isinstance(m, nn.LayerNorm):
+            nn.init.ones_(m.weight)
+            _zero_bias(m)
+        elif isinstance(m, nn.Linear):
+            limit = math.sqrt(3.0 * given_scale / _fan_avg(m))
+            nn.init.uniform_(m.weight, -limit, limit)
+            _zero_bias(m)
 
     return f
