@@ -58,6 +58,7 @@ class BaseAgent:
     """Shared train/deploy/collect_data/batch_evaluate plumbing."""
 
     policy_type: str = None
+    example_algo: str = None  # ding/config/example/<dir> with tuned presets
     on_policy_pipeline: bool = False
     default_policy_cfg: dict = {}
 
@@ -69,23 +70,32 @@ class BaseAgent:
         cfg: Optional[dict] = None,
         policy_state_dict: Optional[Union[str, dict]] = None,
     ):
-        assert env_id in ENV_PRESETS, f"unknown env preset {env_id}; available: {list(ENV_PRESETS)}"
-        preset = copy.deepcopy(ENV_PRESETS[env_id])
         self.env_id = env_id
         self.seed = seed
         self.exp_name = exp_name or f"{env_id}-{self.policy_type}"
-        main = EasyDict(dict(
-            exp_name=self.exp_name,
-            env=preset['env'],
-            policy=deep_merge_dicts(
-                dict(model=dict(obs_shape=preset['obs_shape'], action_shape=preset['action_shape'])),
-                copy.deepcopy(self.default_policy_cfg)
-            ),
-        ))
+        example = None
+        if self.example_algo is not None:
+            from ding.config.example import get_example_config
+            example = get_example_config(self.example_algo, env_id)
+        if example is not None:
+            # tuned per-env preset from ding/config/example/<ALGO>/
+            main = example
+            main.exp_name = self.exp_name
+        else:
+            assert env_id in ENV_PRESETS, f"unknown env preset {env_id}; available: {list(ENV_PRESETS)}"
+            preset = copy.deepcopy(ENV_PRESETS[env_id])
+            main = EasyDict(dict(
+                exp_name=self.exp_name,
+                env=preset['env'],
+                policy=deep_merge_dicts(
+                    dict(model=dict(obs_shape=preset['obs_shape'], action_shape=preset['action_shape'])),
+                    copy.deepcopy(self.default_policy_cfg)
+                ),
+            ))
         if cfg:
             main = EasyDict(deep_merge_dicts(main, cfg))
         create = EasyDict(dict(
-            env=dict(type=preset['env']['type'], import_names=preset['env']['import_names']),
+            env=dict(type=main.env['type'], import_names=main.env['import_names']),
             env_manager=dict(type='base'),
             policy=dict(type=self.policy_type),
         ))
@@ -182,6 +192,7 @@ class BaseAgent:
 
 class DQNAgent(BaseAgent):
     policy_type = 'dqn'
+    example_algo = 'DQN'
     default_policy_cfg = dict(
         nstep=3, discount_factor=0.97,
         learn=dict(update_per_collect=5, batch_size=64, learning_rate=1e-3),
@@ -194,6 +205,7 @@ class DQNAgent(BaseAgent):
 class PPOF(BaseAgent):
     """PPO-of-fans: the simplified high-level PPO (on-policy pipeline)."""
     policy_type = 'ppo'
+    example_algo = 'PPOF'
     on_policy_pipeline = True
     default_policy_cfg = dict(
         action_space='discrete',
@@ -205,6 +217,7 @@ class PPOF(BaseAgent):
 
 class PPOOffPolicyAgent(BaseAgent):
     policy_type = 'ppo_offpolicy'
+    example_algo = 'PPOOffPolicy'
     default_policy_cfg = dict(
         model=dict(action_space='discrete'),
         learn=dict(update_per_collect=4, batch_size=64, epoch_per_collect=1),
@@ -215,6 +228,7 @@ class PPOOffPolicyAgent(BaseAgent):
 
 class A2CAgent(BaseAgent):
     policy_type = 'a2c'
+    example_algo = 'A2C'
     on_policy_pipeline = True
     default_policy_cfg = dict(
         learn=dict(batch_size=64, learning_rate=1e-3),
@@ -224,6 +238,7 @@ class A2CAgent(BaseAgent):
 
 class C51Agent(BaseAgent):
     policy_type = 'c51'
+    example_algo = 'C51'
     default_policy_cfg = dict(
         nstep=3,
         model=dict(v_min=-10, v_max=10, n_atom=51),
@@ -236,6 +251,7 @@ class C51Agent(BaseAgent):
 
 class SACAgent(BaseAgent):
     policy_type = 'sac'
+    example_algo = 'SAC'
     default_policy_cfg = dict(
         random_collect_size=100,
         model=dict(action_space='reparameterization', twin_critic=True),
@@ -247,6 +263,7 @@ class SACAgent(BaseAgent):
 
 class DDPGAgent(BaseAgent):
     policy_type = 'ddpg'
+    example_algo = 'DDPG'
     default_policy_cfg = dict(
         random_collect_size=100,
         model=dict(action_space='regression'),
@@ -258,16 +275,19 @@ class DDPGAgent(BaseAgent):
 
 class TD3Agent(BaseAgent):
     policy_type = 'td3'
+    example_algo = 'TD3'
     default_policy_cfg = DDPGAgent.default_policy_cfg
 
 
 class SQLAgent(BaseAgent):
     policy_type = 'sql'
+    example_algo = 'SQL'
     default_policy_cfg = DQNAgent.default_policy_cfg
 
 
 class PGAgent(BaseAgent):
     policy_type = 'pg'
+    example_algo = 'PG'
     on_policy_pipeline = True
     default_policy_cfg = dict(
         learn=dict(batch_size=64, learning_rate=1e-3),

@@ -1,0 +1,26 @@
+"""Tuned PPOOffPolicy preset for CartPole-v0 (reference
+ding/config/example/PPOOffPolicy/gym_cartpole_v0.py)."""
+from ding.utils import EasyDict
+
+cfg = EasyDict(dict(
+    exp_name='CartPole-v0-PPOOffPolicy',
+    seed=0,
+    env=dict(
+        type='cartpole',
+        import_names=['dizoo.classic_control.cartpole.envs.cartpole_env'],
+        collector_env_num=8,
+        evaluator_env_num=8,
+        n_evaluator_episode=8,
+        stop_value=195,
+        
+    ),
+    policy=dict(
+        cuda=True,
+        action_space='discrete',
+        model=dict(obs_shape=4, action_shape=2, action_space='discrete'),
+        learn=dict(update_per_collect=10, batch_size=320, learning_rate=3e-4, value_weight=0.5,
+                   entropy_weight=0.01, clip_ratio=0.2, adv_norm=True),
+        collect=dict(n_sample=3200, unroll_len=1, discount_factor=0.99, gae_lambda=0.95),
+        other=dict(replay_buffer=dict(replay_buffer_size=10000)),
+    ),
+))

@@ -26,3 +26,29 @@ def test_sac_agent(tmp_path):
     agent.train(step=200, max_train_iter=2)
     ret = agent.deploy(max_episode_steps=50)
     assert isinstance(ret.eval_value, float)
+
+
+def test_bonus_example_presets():
+    """Agents pick up tuned presets from ding/config/example/<ALGO>/."""
+    from ding.config.example import get_example_config, list_example_configs
+    all_presets = list_example_configs()
+    assert len(all_presets) >= 25, all_presets
+    cfg = get_example_config('DQN', 'LunarLander-v2')
+    assert cfg.env.stop_value == 200 and cfg.policy.model.obs_shape == 8
+    assert get_example_config('DQN', 'NoSuchEnv-v0') is None
+
+    from ding.bonus import DQNAgent
+    agent = DQNAgent(env_id='LunarLander-v2', exp_name='exp/test_bonus_ll_dqn')
+    assert agent.main_config.policy.model.action_shape == 4
+    assert agent.create_config.env.type == 'lunarlander'
+    # 1-iter training still works through the preset
+    agent.train(max_train_iter=1, collector_env_num=2, evaluator_env_num=1)
+
+
+def test_bonus_sac_preset_smoke():
+    from ding.bonus import SACAgent
+    agent = SACAgent(env_id='LunarLanderContinuous-v2', exp_name='exp/test_bonus_llc_sac',
+                     cfg=dict(policy=dict(cuda=False, random_collect_size=16,
+                                          learn=dict(batch_size=8),
+                                          other=dict(replay_buffer=dict(replay_buffer_size=1000)))))
+    agent.train(max_train_iter=1, collector_env_num=2, evaluator_env_num=1)
