@@ -61,10 +61,93 @@ def offline_logger(train_show_freq: int = 100) -> Callable:
     return _logger
 
 
-def wandb_online_logger(*args, **kwargs) -> Callable:
-    """wandb unavailable offline: metrics go to the JSONL writer instead."""
-    return online_logger()
+def _wandb_or_none():
+    try:
+        import wandb
+        return wandb
+    except ImportError:
+        return None
 
 
-def wandb_offline_logger(*args, **kwargs) -> Callable:
-    return offline_logger()
+def wandb_online_logger(
+    record_path: str = None,
+    cfg=None,
+    exp_config=None,
+    metric_list: list = None,
+    env=None,
+    model=None,
+    anonymous: bool = False,
+    project_name: str = 'di-engine-amd',
+    run_name: str = None,
+    wandb_sweep: bool = False,
+) -> Callable:
+    """Per-iteration wandb logging of train losses + eval return (reference
+    ding/framework/middleware/functional/logger.py:157 wandb_online_logger).
+    Falls back to the JSONL DistributedWriter lane when wandb is not
+    importable (offline images)."""
+    wandb = _wandb_or_none()
+    if wandb is None:
+        return online_logger()
+    if wandb.run is None:
+        wandb.init(
+            project=project_name, name=run_name, anonymous='allow' if anonymous else None,
+            config=None if exp_config is None else dict(exp_config), reinit=True
+        )
+    if model is not None and cfg is not None and getattr(cfg, 'gradient_logger', False):
+        wandb.watch(model)
+    fallback = online_logger()
+
+    def _logger(ctx: "OnlineRLContext"):
+        fallback(ctx)
+        payload = {'env_step': ctx.env_step, 'train_iter': ctx.train_iter}
+        outputs = ctx.train_output if isinstance(ctx.train_output, list) else (
+            [ctx.train_output] if isinstance(ctx.train_output, dict) else []
+        )
+        for out in outputs:
+            for k, v in out.items():
+                if np.isscalar(v) or (hasattr(v, 'ndim') and getattr(v, 'ndim', 1) == 0):
+                    if metric_list is None or k in metric_list:
+                        payload[f'train/{k}'] = float(v)
+        if ctx.eval_value is not None and not np.isinf(ctx.eval_value):
+            payload['eval/episode_return'] = float(ctx.eval_value)
+        wandb.log(payload, step=ctx.env_step)
+
+    return _logger
+
+
+def wandb_offline_logger(
+    record_path: str = None,
+    cfg=None,
+    exp_config=None,
+    metric_list: list = None,
+    env=None,
+    model=None,
+    anonymous: bool = False,
+    project_name: str = 'di-engine-amd',
+    run_name: str = None,
+    **kwargs,
+) -> Callable:
+    """Offline-RL wandb logging (reference logger.py:402); JSONL fallback
+    when wandb is absent."""
+    wandb = _wandb_or_none()
+    if wandb is None:
+        return offline_logger()
+    if wandb.run is None:
+        wandb.init(
+            project=project_name, name=run_name, anonymous='allow' if anonymous else None,
+            config=None if exp_config is None else dict(exp_config), reinit=True
+        )
+    fallback = offline_logger()
+
+    def _logger(ctx: "OfflineRLContext"):
+        fallback(ctx)
+        payload = {'train_iter': ctx.train_iter}
+        if isinstance(ctx.train_output, dict):
+            for k, v in ctx.train_output.items():
+                if np.isscalar(v) and (metric_list is None or k in metric_list):
+                    payload[f'train/{k}'] = float(v)
+        if ctx.eval_value is not None and not np.isinf(ctx.eval_value):
+            payload['eval/episode_return'] = float(ctx.eval_value)
+        wandb.log(payload, step=ctx.train_iter)
+
+    return _logger
