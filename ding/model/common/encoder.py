@@ -48,7 +48,29 @@ class UnfoldConv2d(nn.Module):
         return out.reshape(B, self.conv.out_channels, Ho, Wo)
 
 
-class StemConv2d(nn.Conv2d):
+class NativeWrwConv2d(nn.Conv2d):
+    """nn.Conv2d whose backward-weights runs the hand-written NHWC HIP
+    kernel (ding/ops/csrc/wrw_ops.hip) on GPU fp32 channels_last inputs —
+    MIOpen resolves NHWC fp32 wrw for the Atari shapes to a naive
+    fp64-accumulate solver. Forward and backward-data stay on MIOpen's
+    tuned solvers; state dict matches nn.Conv2d. Disable with
+    DING_NATIVE_WRW=0."""
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        import os
+        if (
+            os.environ.get('DING_NATIVE_WRW', '1') not in ('0', 'false')
+            and x.is_cuda and x.dtype == torch.float32 and torch.is_grad_enabled()
+            and self.padding == (0, 0) and self.dilation == (1, 1) and self.groups == 1
+            and x.is_contiguous(memory_format=torch.channels_last)
+        ):
+            from ding.ops import dispatch as _dispatch
+            if _dispatch.use_hip_autograd(x):
+                return _dispatch.wrw_conv2d(x, self.weight, self.bias, self.stride)
+        return super().forward(x)
+
+
+class StemConv2d(NativeWrwConv2d):
     """Drop-in nn.Conv2d for the Atari stem (8x8, stride 4, pad 0) that
     routes GPU fp32 forwards through the direct HIP kernel
     (ding/ops/csrc/conv_ops.hip) — rocprof showed MIOpen's tuned choice for
@@ -106,7 +128,10 @@ class ConvEncoder(nn.Module):
                 layers.append(build_activation(activation))
             else:
                 layers.append(
-                    conv2d_block(in_c, hidden_size_list[i], k, s, p, activation=activation, norm_type=norm_type)
+                    conv2d_block(
+                        in_c, hidden_size_list[i], k, s, p, activation=activation, norm_type=norm_type,
+                        conv_cls=NativeWrwConv2d if p == 0 else None
+                    )
                 )
             in_c = hidden_size_list[i]
         layers.append(nn.Flatten())

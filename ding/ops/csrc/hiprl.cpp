@@ -24,6 +24,7 @@ std::vector<torch::Tensor> lstm_cell_bwd(
     torch::Tensor c_in, torch::Tensor c_out, torch::Tensor rstd
 );
 torch::Tensor stem_conv_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias);
+torch::Tensor conv_wrw_nhwc(torch::Tensor x, torch::Tensor dy, int64_t K, int64_t S);
 torch::Tensor stem_conv_wrw(torch::Tensor x, torch::Tensor dy, int64_t O);
 std::vector<torch::Tensor> q_nstep_fwd(
     torch::Tensor q, torch::Tensor next_n_q, torch::Tensor action, torch::Tensor next_action,
@@ -58,6 +59,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("stem_conv_wrw", &stem_conv_wrw, "direct 8x8s4 stem conv weight grad");
     m.def("lstm_cell_fwd", &lstm_cell_fwd, "fused LN-LSTM cell forward");
     m.def("lstm_cell_bwd", &lstm_cell_bwd, "fused LN-LSTM cell backward");
+    m.def("conv_wrw_nhwc", &conv_wrw_nhwc, "NHWC conv2d backward-weights (fp32, register-tiled)");
     m.def("vtrace_fwd", &vtrace_fwd, "fused v-trace loss forward");
     m.def("vtrace_bwd", &vtrace_bwd, "fused v-trace loss backward");
 }

@@ -307,3 +307,34 @@ class _FusedLSTMCell(torch.autograd.Function):
 def fused_lstm_cell(gxn, gh_raw, gamma, beta, bias, c_in):
     """Returns (h', c')."""
     return _FusedLSTMCell.apply(gxn, gh_raw, gamma, beta, bias, c_in)
+
+
+class _WrwConv2dFn(torch.autograd.Function):
+    """conv2d with MIOpen fwd / bwd-data but the HIP NHWC wrw kernel for
+    dW — MIOpen resolves NHWC fp32 wrw for the Atari shapes to a naive
+    fp64-accumulate solver (profiles/impala_nhwc_kernel_stats_r01.csv)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, stride):
+        y = torch.nn.functional.conv2d(x, weight, bias, stride)
+        ctx.save_for_backward(x, weight)
+        ctx.stride = stride
+        ctx.has_bias = bias is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = _load()
+        x, weight = ctx.saved_tensors
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        dx = None
+        if ctx.needs_input_grad[0]:
+            dx = torch.nn.grad.conv2d_input(x.shape, weight, dy, stride=ctx.stride)
+        dw = ext.conv_wrw_nhwc(x, dy, weight.shape[-1], ctx.stride[0] if isinstance(ctx.stride, (tuple, list)) else ctx.stride)
+        db = dy.sum(dim=(0, 2, 3)) if ctx.has_bias else None
+        return dx, dw, db, None
+
+
+def wrw_conv2d(x, weight, bias, stride):
+    """conv2d(x, w, b, stride) with the hand-written NHWC wrw backward."""
+    return _WrwConv2dFn.apply(x, weight, bias, stride)

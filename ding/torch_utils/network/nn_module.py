@@ -109,16 +109,18 @@ def conv2d_block(
     norm_type: Optional[str] = None,
     num_groups_for_gn: int = 1,
     bias: bool = True,
+    conv_cls: type = None,
 ) -> nn.Sequential:
+    conv_cls = conv_cls or nn.Conv2d
     layers = []
     if pad_type == "zero":
         layers.append(
-            nn.Conv2d(in_channels, out_channels, kernel_size, stride, padding, dilation, groups, bias=bias)
+            conv_cls(in_channels, out_channels, kernel_size, stride, padding, dilation, groups, bias=bias)
         )
     elif pad_type in ("reflect", "replicate"):
         pad_cls = nn.ReflectionPad2d if pad_type == "reflect" else nn.ReplicationPad2d
         layers.append(pad_cls(padding))
-        layers.append(nn.Conv2d(in_channels, out_channels, kernel_size, stride, 0, dilation, groups, bias=bias))
+        layers.append(conv_cls(in_channels, out_channels, kernel_size, stride, 0, dilation, groups, bias=bias))
     else:
         raise KeyError(pad_type)
     if norm_type is not None:

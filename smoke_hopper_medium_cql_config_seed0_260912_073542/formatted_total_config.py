@@ -1,0 +1,87 @@
+exp_config = {
+    "exp_name": "smoke_hopper_medium_cql_config_seed0_260912_073542",
+    "seed": 0,
+    "env": {
+        "manager": {
+            "episode_num": float('inf'),
+            "max_retry": 1,
+            "retry_type": "reset",
+            "auto_reset": True,
+            "step_timeout": None,
+            "reset_timeout": None,
+            "retry_waiting_time": 0.1,
+            "shared_memory": True,
+            "copy_on_get": True,
+            "type": "base"
+        },
+        "type": "d4rl",
+        "env_id": "hopper-medium-v2",
+        "collector_env_num": 2,
+        "evaluator_env_num": 1,
+        "n_evaluator_episode": 1,
+        "stop_value": 6000,
+        "use_act_scale": True,
+        "max_step": 30,
+        "import_names": [
+            "dizoo.d4rl.envs.d4rl_env"
+        ]
+    },
+    "policy": {
+        "on_policy": False,
+        "cuda": False,
+        "multi_gpu": False,
+        "bp_update_sync": True,
+        "traj_len_inf": False,
+        "model": {
+            "twin_critic": True,
+            "action_space": "reparameterization",
+            "obs_shape": 11,
+            "action_shape": 3,
+            "actor_head_hidden_size": 256,
+            "critic_head_hidden_size": 256
+        },
+        "type": "cql_command",
+        "multi_agent": False,
+        "priority": False,
+        "priority_IS_weight": False,
+        "random_collect_size": 10000,
+        "transition_with_policy_data": True,
+        "learn": {
+            "update_per_collect": 1,
+            "batch_size": 8,
+            "learning_rate_q": 0.0003,
+            "learning_rate_policy": 0.0001,
+            "learning_rate_alpha": 0.0001,
+            "target_theta": 0.005,
+            "discount_factor": 0.99,
+            "alpha": 0.2,
+            "auto_alpha": True,
+            "log_space": True,
+            "ignore_done": False,
+            "target_entropy": None,
+            "min_q_weight": 5.0,
+            "with_lagrange": False,
+            "lagrange_thresh": -1,
+            "num_actions": 10
+        },
+        "collect": {
+            "unroll_len": 1,
+            "data_type": "hdf5",
+            "data_path": "/tmp/pytest-of-root/pytest-94/test_dizoo_config_smoke_dizoo_35/hopper-medium-v2.npz",
+            "normalize_states": True
+        },
+        "eval": {
+            "evaluator": {
+                "eval_freq": 1000,
+                "stop_value": 6000,
+                "n_episode": 1
+            }
+        },
+        "other": {
+            "replay_buffer": {
+                "replay_buffer_size": 1000000
+            }
+        },
+        "cfg_type": "CQLPolicyCommandDict"
+    }
+}

@@ -1,0 +1,98 @@
+exp_config = {
+    "exp_name": "smoke_pomdp_ppo_seed0_260912_073536",
+    "seed": 0,
+    "env": {
+        "manager": {
+            "episode_num": float('inf'),
+            "max_retry": 1,
+            "retry_type": "reset",
+            "auto_reset": True,
+            "step_timeout": None,
+            "reset_timeout": None,
+            "retry_waiting_time": 0.1,
+            "shared_memory": True,
+            "copy_on_get": True,
+            "type": "base"
+        },
+        "type": "pomdp",
+        "collector_env_num": 2,
+        "evaluator_env_num": 1,
+        "n_evaluator_episode": 1,
+        "stop_value": 20,
+        "env_id": "Pong-ramNoFrameskip-v4",
+        "frame_stack": 4,
+        "warp_frame": False,
+        "use_ram": True,
+        "pomdp": {
+            "noise_scale": 0.01,
+            "zero_p": 0.2,
+            "reward_noise": 0.01,
+            "duplicate_p": 0.2
+        },
+        "max_step": 30,
+        "import_names": [
+            "dizoo.pomdp.envs.pomdp_env"
+        ]
+    },
+    "policy": {
+        "on_policy": False,
+        "cuda": False,
+        "multi_gpu": False,
+        "bp_update_sync": True,
+        "traj_len_inf": False,
+        "model": {
+            "obs_shape": [
+                512
+            ],
+            "action_shape": 6,
+            "action_space": "discrete",
+            "encoder_hidden_size_list": [
+                128,
+                128,
+                64
+            ]
+        },
+        "type": "ppo_offpolicy_command",
+        "priority": False,
+        "priority_IS_weight": False,
+        "recompute_adv": False,
+        "action_space": "discrete",
+        "nstep_return": False,
+        "multi_agent": False,
+        "transition_with_policy_data": True,
+        "learn": {
+            "epoch_per_collect": 10,
+            "batch_size": 8,
+            "learning_rate": 0.0001,
+            "value_weight": 0.5,
+            "entropy_weight": 0.01,
+            "clip_ratio": 0.1,
+            "adv_norm": True,
+            "value_norm": True,
+            "ppo_param_init": True,
+            "grad_clip_type": "clip_norm",
+            "grad_clip_value": 0.5,
+            "ignore_done": False,
+            "update_per_collect": 1
+        },
+        "collect": {
+            "unroll_len": 1,
+            "discount_factor": 0.99,
+            "gae_lambda": 0.95,
+            "n_sample": 16
+        },
+        "eval": {
+            "evaluator": {
+                "eval_freq": 100,
+                "stop_value": 20,
+                "n_episode": 1
+            }
+        },
+        "other": {
+            "replay_buffer": {
+                "replay_buffer_size": 1000
+            }
+        },
+        "cfg_type": "PPOOffPolicyCommandDict"
+    }
+}
