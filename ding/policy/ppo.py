@@ -105,6 +105,10 @@ class PPOPolicy(Policy):
         self._gamma = self._cfg.collect.discount_factor
         self._gae_lambda = self._cfg.collect.gae_lambda
         self._recompute_adv = self._cfg.recompute_adv
+        # opt-in bf16 learner lane: model fwd (convs/GEMMs) runs under
+        # autocast-bf16 with fp32 master weights and fp32 loss math — bf16
+        # needs no loss scaling. The hipGraph path stays fp32-only.
+        self._bf16 = self._cfg.learn.get('bf16', False)
         # hipGraph capture of the minibatch step (MI355X: the step is
         # launch-bound — see ding/torch_utils/hip_graph.py). Single-process
         # only: the bucketed DDP reducer's backward hooks are not replayed

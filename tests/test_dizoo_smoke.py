@@ -12,7 +12,7 @@ def _shrink(main_cfg, create_cfg):
     """Scale a tuned config down to smoke size (tiny envs/batches, no eval)."""
     m = copy.deepcopy(main_cfg)
     c = copy.deepcopy(create_cfg)
-    m.exp_name = 'smoke_' + m.exp_name
+    m.exp_name = 'exp/smoke_' + m.exp_name
     m.env.collector_env_num = 2
     m.env.evaluator_env_num = 1
     m.env.n_evaluator_episode = 1
