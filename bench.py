@@ -46,7 +46,7 @@ def _maybe_channels_last(policy):
     return policy
 
 
-def build_ppo_policy(device: str, multi_gpu: bool):
+def build_ppo_policy(device: str, multi_gpu: bool, bf16: bool = False):
     from ding.policy import PPOPolicy
     from ding.utils import EasyDict, deep_merge_dicts
 
@@ -76,9 +76,11 @@ def build_ppo_policy(device: str, multi_gpu: bool):
             ignore_done=False,
             grad_clip_type='clip_norm',
             grad_clip_value=0.5,
+            bf16=bf16,
             # hipGraph-capture the minibatch fwd+loss+bwd (launch-bound step;
-            # single-process only — the policy ignores this under multi_gpu)
-            cuda_graph=os.environ.get('DING_PPO_GRAPH', '1') not in ('0', 'false'),
+            # single-process only — the policy ignores this under multi_gpu;
+            # the bf16 lane takes the eager autocast path)
+            cuda_graph=(not bf16) and os.environ.get('DING_PPO_GRAPH', '1') not in ('0', 'false'),
             channels_last=_channels_last_on(),
         ),
         collect=dict(n_sample=3200, unroll_len=1, discount_factor=0.99, gae_lambda=0.95),
@@ -197,6 +199,7 @@ def main():
     parser.add_argument('--warmup', type=int, default=2)
     parser.add_argument('--workload', type=str, default='ppo', choices=['ppo', 'impala'])
     parser.add_argument('--n-sample', type=int, default=3200)
+    parser.add_argument('--dtype', type=str, default='fp32', choices=['fp32', 'bf16'])
     args = parser.parse_args()
 
     world_size = int(os.environ.get('WORLD_SIZE', '1'))
@@ -219,7 +222,7 @@ def main():
     torch.manual_seed(1234 + rank)
 
     if args.workload == 'ppo':
-        policy = build_ppo_policy(device, multi_gpu=distributed)
+        policy = build_ppo_policy(device, multi_gpu=distributed, bf16=args.dtype == 'bf16')
         step_fn = lambda: ppo_step(policy, device, args.n_sample)
         model_name = 'pong_ppo(conv[64,64,128] 4x84x84)'
         config = {
@@ -274,7 +277,7 @@ def main():
             'higher_is_better': True,
             'scaling': 'weak',
             'vs_baseline': None,
-            'dtype': 'fp32',
+            'dtype': args.dtype,
             'data': 'synthetic',
             'config': config,
         }))

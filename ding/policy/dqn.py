@@ -91,10 +91,17 @@ class DQNPolicy(Policy):
             data = to_device(data, self._device)
         self._learn_model.train()
         self._target_model.train()
-        with torch.no_grad():
-            target_q_value = self._target_model.forward(data['next_obs'])['logit']
+        # opt-in bf16 lane: net fwd in autocast-bf16 (fp32 master weights,
+        # fp32 TD/loss math, no loss scaling)
+        import contextlib
+        amp = torch.autocast(data['obs'].device.type, dtype=torch.bfloat16) \
+            if self._cfg.learn.get('bf16', False) else contextlib.nullcontext()
+        with torch.no_grad(), amp:
+            target_q_value = self._target_model.forward(data['next_obs'])['logit'].float()
             target_q_action = self._learn_model.forward(data['next_obs'])['action']  # double DQN
-        q_value = self._learn_model.forward(data['obs'])['logit']
+        with amp:
+            q_value = self._learn_model.forward(data['obs'])['logit']
+        q_value = q_value.float()
         value_gamma = data.get('value_gamma')
         td_data = q_nstep_td_data(
             q_value, target_q_value, data['action'], target_q_action, data['reward'], data['done'], data['weight']

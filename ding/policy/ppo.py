@@ -123,6 +123,14 @@ class PPOPolicy(Policy):
         self._values_graph = None
         self._learn_model.reset()
 
+    def _amp_ctx(self, ref: torch.Tensor):
+        """autocast-bf16 context for the opt-in bf16 lane (nullcontext when
+        the lane is off); fp32 master weights, no loss scaling."""
+        import contextlib
+        if not self._bf16 or not isinstance(ref, torch.Tensor):
+            return contextlib.nullcontext()
+        return torch.autocast(ref.device.type, dtype=torch.bfloat16)
+
     def _graphed_values(self, both: torch.Tensor, chunk: int, fresh: bool = True) -> torch.Tensor:
         """hipGraph-captured no-grad chunked critic pass for recompute-adv.
         The [obs; next_obs] tensor is identical across the epoch loop, so
@@ -226,14 +234,19 @@ class PPOPolicy(Policy):
                         import os as _os
                         _chunk = int(_os.environ.get('DING_PPO_VALUE_CHUNK', 0)) or \
                             max(int(self._cfg.learn.batch_size), 3200)
-                        if self._cuda_graph and both.is_cuda and both.dtype == torch.float32:
+                        if self._cuda_graph and both.is_cuda and both.dtype == torch.float32 \
+                                and not self._bf16:
                             values = self._graphed_values(both, _chunk, fresh=(epoch == 0))
                         else:
                             chunks = torch.split(both, _chunk, dim=0)
-                            values = torch.cat(
-                                [self._learn_model.forward(c, mode='compute_critic')['value'] for c in chunks],
-                                dim=0
-                            )
+                            with self._amp_ctx(both):
+                                values = torch.cat(
+                                    [
+                                        self._learn_model.forward(c, mode='compute_critic')['value']
+                                        for c in chunks
+                                    ],
+                                    dim=0
+                                ).float()
                         value, next_value = values.chunk(2, dim=0)
                     else:
                         value = self._learn_model.forward(data['obs'], mode='compute_critic')['value']
@@ -263,7 +276,7 @@ class PPOPolicy(Policy):
             graph_infos = []
             for batch in split_data_generator(data, self._cfg.learn.batch_size, shuffle=True):
                 if (
-                    self._cuda_graph and self._action_space == 'discrete'
+                    self._cuda_graph and self._action_space == 'discrete' and not self._bf16
                     and isinstance(batch['obs'], torch.Tensor) and batch['obs'].is_cuda
                     and batch.get('weight') is None and batch['obs'].dtype == torch.float32
                 ):
@@ -274,7 +287,14 @@ class PPOPolicy(Policy):
                     # at the end of the epoch — avoids a device sync per minibatch
                     graph_infos.append({k: v.clone() for k, v in out.items()})
                     continue
-                output = self._learn_model.forward(batch['obs'], mode='compute_actor_critic')
+                with self._amp_ctx(batch['adv']):
+                    output = self._learn_model.forward(batch['obs'], mode='compute_actor_critic')
+                if self._bf16:  # loss math stays fp32
+                    output = {
+                        k: (v.float() if isinstance(v, torch.Tensor) else
+                            {kk: vv.float() for kk, vv in v.items()} if isinstance(v, dict) else v)
+                        for k, v in output.items()
+                    }
                 adv = batch['adv']
                 if self._adv_norm:
                     adv = (adv - adv.mean()) / (adv.std() + 1e-8)

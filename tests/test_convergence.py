@@ -74,3 +74,44 @@ def test_pendulum_sac_improves():
     assert ckpts
     value = eval_entry((main, create), seed=0, load_path=sorted(ckpts)[-1])
     assert value > -900, f"best-ckpt eval {value} <= -900 (random-level)"
+
+
+def test_cartpole_ppo_bf16_converges():
+    """bf16 learner lane (autocast-bf16 fwd, fp32 master weights) reaches
+    the same convergence gate as fp32."""
+    from ding.entry import serial_pipeline_onpolicy
+    main, create = cartpole_cfg('ppo', extra_policy=dict(
+        action_space='discrete', recompute_adv=True,
+        model=dict(obs_shape=4, action_shape=2, encoder_hidden_size_list=[64, 64]),
+        learn=dict(epoch_per_collect=2, batch_size=64, learning_rate=3e-4, bf16=True),
+        collect=dict(n_sample=256, discount_factor=0.99, gae_lambda=0.95),
+        eval=dict(evaluator=dict(eval_freq=20)),
+    ))
+    main.env.stop_value = 195
+    main.exp_name = 'exp/conv_ppo_bf16'
+    serial_pipeline_onpolicy((main, create), seed=0, max_env_step=150000)
+    from ding.entry import eval as eval_entry
+    import glob
+    ckpts = glob.glob(f'{main.exp_name}*/ckpt/ckpt_best.pth.tar')
+    assert ckpts
+    value = eval_entry((main, create), seed=0, load_path=sorted(ckpts)[-1])
+    assert value >= 100, f"bf16 best-ckpt eval {value} < 100"
+
+
+def test_cartpole_dqn_bf16_learn_step():
+    """DQN bf16 lane: one learn step produces finite fp32 grads/updates."""
+    import torch
+    from ding.policy import DQNPolicy
+    from ding.utils import EasyDict, deep_merge_dicts
+    cfg = EasyDict(deep_merge_dicts(DQNPolicy.default_config(), EasyDict(dict(
+        cuda=False, nstep=1,
+        model=dict(obs_shape=4, action_shape=2, encoder_hidden_size_list=[32, 32]),
+        learn=dict(batch_size=8, update_per_collect=1, learning_rate=1e-3, bf16=True),
+    ))))
+    pol = DQNPolicy(cfg, enable_field=['learn'])
+    data = [dict(obs=torch.randn(4), next_obs=torch.randn(4), action=torch.tensor([0]),
+                 reward=torch.tensor([1.0]), done=False) for _ in range(8)]
+    out = pol._forward_learn(data)
+    assert all(p.dtype == torch.float32 for p in pol._model.parameters()), "master weights stay fp32"
+    import math
+    assert math.isfinite(out['total_loss'])
