@@ -8,3 +8,4 @@ from .storage_loader import StorageLoader, FileStorageLoader
 from .model_loader import ModelLoader, FileModelLoader
 from .level_replay import LevelSampler
 from .buffer import GPUPrioritizedBuffer
+from .tensor_shipper import TrajectoryShipper
