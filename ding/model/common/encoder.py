@@ -50,16 +50,21 @@ class UnfoldConv2d(nn.Module):
 
 class NativeWrwConv2d(nn.Conv2d):
     """nn.Conv2d whose backward-weights runs the hand-written NHWC HIP
-    kernel (ding/ops/csrc/wrw_ops.hip) on GPU fp32 channels_last inputs —
-    MIOpen resolves NHWC fp32 wrw for the Atari shapes to a naive
-    fp64-accumulate solver. Forward and backward-data stay on MIOpen's
-    tuned solvers; state dict matches nn.Conv2d. Disable with
-    DING_NATIVE_WRW=0."""
+    kernel (ding/ops/csrc/wrw_ops.hip) on GPU fp32 channels_last inputs.
+
+    OPT-IN (DING_NATIVE_WRW=1): measured on MI355X the hand-written wrw is
+    numerically exact but 2-4x slower than MIOpen's tuned implicit-GEMM
+    picks at the Atari shapes (hip 0.16-0.36 ms vs miopen 0.04-0.09 ms,
+    tests/test_ops_gpu.py::test_conv_wrw_timing), and the PPO bench
+    regressed 17.8k -> 13.4k samples/s with it on. The round-1 naive-fp64
+    wrw solver only appears inside MIOpen find-mode contexts and was
+    already off the critical path (profiles/README.md), so MIOpen stays
+    the default; the kernel remains as the guaranteed-fp32 fallback."""
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         import os
         if (
-            os.environ.get('DING_NATIVE_WRW', '1') not in ('0', 'false')
+            os.environ.get('DING_NATIVE_WRW', '0') in ('1', 'true')
             and x.is_cuda and x.dtype == torch.float32 and torch.is_grad_enabled()
             and self.padding == (0, 0) and self.dilation == (1, 1) and self.groups == 1
             and x.is_contiguous(memory_format=torch.channels_last)
