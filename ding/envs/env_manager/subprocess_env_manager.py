@@ -157,7 +157,7 @@ class SyncSubprocessEnvManager(BaseEnvManager):
         return payload
 
     def _get_obs(self, env_id: int, payload):
-        if isinstance(payload, tuple) and len(payload) >= 1 and payload[0] == 'shm':
+        if isinstance(payload, tuple) and len(payload) >= 1 and isinstance(payload[0], str) and payload[0] == 'shm':
             return self._shm[env_id].get()
         return payload
 
@@ -184,7 +184,7 @@ class SyncSubprocessEnvManager(BaseEnvManager):
             try:
                 payload = self._recv(env_id, timeout=self._reset_timeout)
                 obs = self._get_obs(env_id, payload)
-                if not (isinstance(obs, tuple) and obs and obs[0] == 'shm'):
+                if not (isinstance(obs, tuple) and obs and isinstance(obs[0], str) and obs[0] == 'shm'):
                     self._ready_obs[env_id] = obs
                 else:
                     self._ready_obs[env_id] = self._shm[env_id].get()
@@ -192,7 +192,13 @@ class SyncSubprocessEnvManager(BaseEnvManager):
                 return
             except BaseException as e:
                 exceptions.append(e)
-                self._renew_env(env_id)
+                if attempt + 1 >= self._max_retry:
+                    break
+                if self._cfg.get('retry_type', 'renew') == 'renew' or isinstance(e, TimeoutError):
+                    # hung/dead worker: replace the subprocess entirely
+                    self._renew_env(env_id)
+                else:
+                    time.sleep(self._cfg.get('retry_waiting_time', 0.1))
                 self._pipes[env_id].send(('reset', self._reset_param[env_id]))
         self._env_states[env_id] = EnvState.ERROR
         self.close()
@@ -204,7 +210,7 @@ class SyncSubprocessEnvManager(BaseEnvManager):
         timesteps = {}
         for env_id in actions:
             payload = self._recv(env_id, timeout=self._step_timeout)
-            if payload[0] == 'shm':
+            if isinstance(payload[0], str) and payload[0] == 'shm':
                 _, reward, done, info = payload
                 obs = self._shm[env_id].get()
             else:
@@ -286,7 +292,7 @@ class AsyncSubprocessEnvManager(SyncSubprocessEnvManager):
             time.sleep(0.001)
 
     def _finish_step(self, env_id: int, payload):
-        if payload[0] == 'shm':
+        if isinstance(payload[0], str) and payload[0] == 'shm':
             _, reward, done, info = payload
             obs = self._shm[env_id].get()
         else:
