@@ -79,8 +79,8 @@ def build_ppo_policy(device: str, multi_gpu: bool, bf16: bool = False):
             bf16=bf16,
             # hipGraph-capture the minibatch fwd+loss+bwd (launch-bound step;
             # single-process only — the policy ignores this under multi_gpu;
-            # the bf16 lane takes the eager autocast path)
-            cuda_graph=(not bf16) and os.environ.get('DING_PPO_GRAPH', '1') not in ('0', 'false'),
+            # under bf16 the autocast weight-casts are captured into the graph)
+            cuda_graph=os.environ.get('DING_PPO_GRAPH', '1') not in ('0', 'false'),
             channels_last=_channels_last_on(),
         ),
         collect=dict(n_sample=3200, unroll_len=1, discount_factor=0.99, gae_lambda=0.95),

@@ -107,7 +107,8 @@ class PPOPolicy(Policy):
         self._recompute_adv = self._cfg.recompute_adv
         # opt-in bf16 learner lane: model fwd (convs/GEMMs) runs under
         # autocast-bf16 with fp32 master weights and fp32 loss math — bf16
-        # needs no loss scaling. The hipGraph path stays fp32-only.
+        # needs no loss scaling. Works inside hipGraph capture too (the
+        # autocast weight-cast kernels are captured and replayed).
         self._bf16 = self._cfg.learn.get('bf16', False)
         # hipGraph capture of the minibatch step (MI355X: the step is
         # launch-bound — see ding/torch_utils/hip_graph.py). Single-process
@@ -141,12 +142,12 @@ class PPOPolicy(Policy):
             from ding.torch_utils.hip_graph import GraphedStep
 
             def fn(inp):
-                with torch.no_grad():
+                with torch.no_grad(), self._amp_ctx(inp['both']):
                     chunks = torch.split(inp['both'], chunk, dim=0)
                     return {
                         'values': torch.cat(
                             [self._learn_model.forward(c, mode='compute_critic')['value'] for c in chunks], dim=0
-                        )
+                        ).float()
                     }
 
             self._values_graph = GraphedStep(fn)
@@ -169,7 +170,10 @@ class PPOPolicy(Policy):
             wv, we = self._value_weight, self._entropy_weight
 
             def step_fn(b):
-                output = self._learn_model.forward(b['obs'], mode='compute_actor_critic')
+                with self._amp_ctx(b['obs']):
+                    output = self._learn_model.forward(b['obs'], mode='compute_actor_critic')
+                if self._bf16:
+                    output = {k: v.float() for k, v in output.items()}
                 adv = b['adv']
                 if self._adv_norm:
                     adv = (adv - adv.mean()) / (adv.std() + 1e-8)
@@ -234,8 +238,7 @@ class PPOPolicy(Policy):
                         import os as _os
                         _chunk = int(_os.environ.get('DING_PPO_VALUE_CHUNK', 0)) or \
                             max(int(self._cfg.learn.batch_size), 3200)
-                        if self._cuda_graph and both.is_cuda and both.dtype == torch.float32 \
-                                and not self._bf16:
+                        if self._cuda_graph and both.is_cuda and both.dtype == torch.float32:
                             values = self._graphed_values(both, _chunk, fresh=(epoch == 0))
                         else:
                             chunks = torch.split(both, _chunk, dim=0)
@@ -276,7 +279,7 @@ class PPOPolicy(Policy):
             graph_infos = []
             for batch in split_data_generator(data, self._cfg.learn.batch_size, shuffle=True):
                 if (
-                    self._cuda_graph and self._action_space == 'discrete' and not self._bf16
+                    self._cuda_graph and self._action_space == 'discrete'
                     and isinstance(batch['obs'], torch.Tensor) and batch['obs'].is_cuda
                     and batch.get('weight') is None and batch['obs'].dtype == torch.float32
                 ):
