@@ -39,3 +39,6 @@ from .misc_helpers import (
 )
 from .misc_helpers import get_ip, get_pid, PropagatingThread, deprecated
 from typing import Sequence as SequenceType  # reference utils/type_helper.py
+from .k8s_helper import (
+    get_operator_server_kwargs, exist_operator_server, pod_exec_command, OperatorServer, OrchestratorLauncher,
+)

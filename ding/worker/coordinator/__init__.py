@@ -1,0 +1,1 @@
+from ding.utils.k8s_helper import OperatorServer

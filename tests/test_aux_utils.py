@@ -176,3 +176,65 @@ def test_wandb_logger_fallback_without_wandb():
     ctx.train_output = {'total_loss': 0.1}
     ctx.eval_value = float('-inf')
     lg(ctx)  # must not raise
+
+
+def test_operator_server_against_stub_orchestrator():
+    """OperatorServer client round-trips replicas API against a local stub
+    DI-orchestrator (flask)."""
+    import threading
+    import time
+    from flask import Flask, jsonify, request as freq
+    from ding.utils import OperatorServer
+    from ding.utils.misc_helpers import find_free_port
+
+    app = Flask('stub_orchestrator')
+    state = {'collectors': [], 'learners': [], 'failed': []}
+
+    @app.route('/v1alpha1/replicas', methods=['GET', 'POST', 'DELETE'])
+    def replicas():
+        data = freq.get_json(silent=True) or {}
+        if freq.method == 'POST':
+            state['collectors'] = [f"cl-{i}" for i in range(int(data.get('collectors', 0)))]
+            state['learners'] = [f"ln-{i}" for i in range(int(data.get('learners', 0)))]
+        if freq.method == 'DELETE':
+            state['collectors'] = state['collectors'][:-int(data['collectors']['replicas']) or None]
+        return jsonify({'code': 0, 'message': 'success',
+                        'data': {'collectors': state['collectors'], 'learners': state['learners']}})
+
+    @app.route('/v1alpha1/replicas/failed', methods=['POST'])
+    def failed():
+        data = freq.get_json(silent=True) or {}
+        state['failed'] = data.get('collectors', []) + data.get('learners', [])
+        return jsonify({'code': 0, 'message': 'success', 'data': state['failed']})
+
+    port = find_free_port()
+    th = threading.Thread(target=lambda: app.run(host='127.0.0.1', port=port), daemon=True)
+    th.start()
+    time.sleep(0.5)
+
+    srv = OperatorServer(host='127.0.0.1', port=port, namespace='di-test', name='coord-0')
+    srv.set_worker_type('coordinator')
+    ok, code, msg, data = srv.post_replicas({'collectors': 2, 'learners': 1})
+    assert ok and len(data['collectors']) == 2 and len(data['learners']) == 1
+    ok, _, _, data = srv.get_replicas()
+    assert ok and data['collectors'] == ['cl-0', 'cl-1']
+    ok, _, _, data = srv.post_replicas_failed(collectors=['cl-1'])
+    assert ok and data == ['cl-1']
+    ok, _, _, data = srv.delete_replicas(n_collectors=1)
+    assert ok and len(data['collectors']) == 1
+
+
+def test_k8s_env_kwargs_and_manifests(tmp_path, monkeypatch):
+    from ding.utils import get_operator_server_kwargs, exist_operator_server, OrchestratorLauncher
+    from ding.utils import EasyDict
+    monkeypatch.setenv('KUBERNETES_SERVER_URL', 'di-server.di-system:8080')
+    monkeypatch.setenv('KUBERNETES_POD_NAMESPACE', 'ns1')
+    monkeypatch.setenv('KUBERNETES_POD_NAME', 'job-coordinator')
+    assert exist_operator_server()
+    kw = get_operator_server_kwargs(EasyDict({}))
+    assert kw == {'api_version': '/v1alpha1', 'namespace': 'ns1', 'name': 'job-coordinator',
+                  'host': 'di-server.di-system', 'port': 8080}
+    launcher = OrchestratorLauncher(namespace='di-system')
+    manifest = launcher.create_manifest(output_path=str(tmp_path / 'orch.yaml'))
+    assert 'di-operator' in manifest and 'di-server' in manifest
+    assert (tmp_path / 'orch.yaml').exists()
