@@ -122,6 +122,7 @@ class PPOPolicy(Policy):
             self._model.to(memory_format=torch.channels_last)
         self._graphed_step = None
         self._values_graph = None
+        self._graph_info_keys = None
         self._learn_model.reset()
 
     def _amp_ctx(self, ref: torch.Tensor):
@@ -288,7 +289,11 @@ class PPOPolicy(Policy):
                     })
                     # static outputs: clone (async) now, convert to floats once
                     # at the end of the epoch — avoids a device sync per minibatch
-                    graph_infos.append({k: v.clone() for k, v in out.items()})
+                    # ONE packed stack per minibatch instead of a clone per
+                    # stat (rocprof: copyBuffer was ~2.7k launches/step)
+                    if not hasattr(self, '_graph_info_keys') or self._graph_info_keys is None:
+                        self._graph_info_keys = list(out.keys())
+                    graph_infos.append(torch.stack([out[k] for k in self._graph_info_keys]))
                     continue
                 with self._amp_ctx(batch['adv']):
                     output = self._learn_model.forward(batch['obs'], mode='compute_actor_critic')
@@ -363,8 +368,10 @@ class PPOPolicy(Policy):
             if graph_infos:
                 # one host sync for the whole epoch's graphed minibatches
                 lr = self._optimizer.defaults['lr']
-                for g in graph_infos:
-                    info = {k: float(v) for k, v in g.items()}
+                packed = torch.stack(graph_infos).cpu()  # [n_minibatch, n_keys]
+                keys = self._graph_info_keys
+                for row in packed:
+                    info = {k: float(v) for k, v in zip(keys, row)}
                     info['cur_lr'] = lr
                     return_infos.append(info)
         return return_infos

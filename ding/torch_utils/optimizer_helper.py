@@ -90,9 +90,22 @@ class Adam(torch.optim.Adam):
         # capturable=True keeps Adam's step counters on-device so the whole
         # optimizer update can live inside a hipGraph capture (MI355X:
         # removes ~30 launches + host work per replayed minibatch)
+        params = list(params)
+        leaves = [p for p in params if isinstance(p, torch.Tensor)]
+        # fused Adam: ONE multi-tensor kernel per step instead of the foreach
+        # chain — rocprof on the PPO bench showed the eager optimizer
+        # (multi_tensor_apply + fills + adds) at ~27 ms/step of the 180 ms
+        # step. GPU-resident float params only; DING_FUSED_ADAM=0 disables.
+        import os as _os
+        use_fused = (
+            _os.environ.get('DING_FUSED_ADAM', '1') not in ('0', 'false') and not capturable
+            and len(leaves) == len(params) and len(leaves) > 0
+            and all(p.is_cuda and p.dtype in (torch.float32, torch.float16, torch.bfloat16) for p in leaves)
+        )
         super().__init__(
             params, lr=lr, betas=betas, eps=eps, weight_decay=weight_decay, amsgrad=amsgrad,
-            capturable=capturable, foreach=True if capturable else None,
+            capturable=capturable, foreach=(True if capturable else None) if not use_fused else None,
+            fused=use_fused or None,
         )
         if grad_clip_type in ('clip_momentum_norm', 'ignore_momentum_norm'):
             for group in self.param_groups:
@@ -110,7 +123,9 @@ class Adam(torch.optim.Adam):
         if t == 'clip_value':
             nn.utils.clip_grad_value_(params, self._clip_value)
         elif t == 'clip_norm':
-            nn.utils.clip_grad_norm_(params, self._clip_value, self._clip_norm_type)
+            # foreach=True: one fused norm + scale instead of a reduce per param
+            _fe = True if (params and all(p.is_cuda for p in params)) else None
+            nn.utils.clip_grad_norm_(params, self._clip_value, self._clip_norm_type, foreach=_fe)
         elif t == 'ignore_value':
             grad_ignore_value(params, self._ignore_value)
         elif t == 'ignore_norm':
@@ -199,7 +214,9 @@ class RMSprop(torch.optim.RMSprop):
         if t == 'clip_value':
             nn.utils.clip_grad_value_(params, self._clip_value)
         elif t == 'clip_norm':
-            nn.utils.clip_grad_norm_(params, self._clip_value, self._clip_norm_type)
+            # foreach=True: one fused norm + scale instead of a reduce per param
+            _fe = True if (params and all(p.is_cuda for p in params)) else None
+            nn.utils.clip_grad_norm_(params, self._clip_value, self._clip_norm_type, foreach=_fe)
         elif t == 'ignore_value':
             grad_ignore_value(params, self._ignore_value)
         elif t == 'ignore_norm':
