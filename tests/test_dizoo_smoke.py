@@ -91,6 +91,7 @@ SMOKE_CONFIGS = [
     ('dizoo.dmc2gym.config.dmc2gym_sac_state_config', 'serial'),
     ('dizoo.dmc2gym.config.dmc2gym_sac_pixel_config', 'serial'),
     ('dizoo.dmc2gym.config.dmc2gym_ppo_config', 'onpolicy'),
+    ('dizoo.cliffwalking.config.cliffwalking_dqn_config', 'serial'),
 ]
 
 
@@ -189,3 +190,59 @@ def test_dmc2gym_pixel_render_tracks_state():
     env2.reset()
     r_bal = float(env2.step(np.array([0.0])).reward[0])
     assert r_bal > 0.5
+
+
+def test_cliffwalking_env_optimal_path():
+    """Greedy optimal policy scores -13 (the stop_value)."""
+    import numpy as np
+    from dizoo.cliffwalking.envs.cliffwalking_env import CliffWalkingEnv
+    env = CliffWalkingEnv({})
+    env.seed(0)
+    env.reset()
+    ret = 0.0
+    for a in [0] + [1] * 11 + [2]:  # up, 11x right, down
+        ts = env.step(np.array([a]))
+        ret += float(ts.reward[0])
+    assert ts.done and abs(ret - (-13.0)) < 1e-6
+
+
+def test_maze_bc_with_bfs_expert():
+    """Maze + value-iteration BFS expert -> one BC learn epoch (the
+    procedure-cloning data path, reference dizoo/maze/config/maze_bc)."""
+    import numpy as np
+    import torch
+    from dizoo.maze.envs.maze_env import MazeEnv
+    from ding.utils.misc_helpers import get_vi_sequence
+    from ding.policy import create_policy
+    from ding.utils import EasyDict, deep_merge_dicts
+    from ding.policy.offline import BehaviourCloningPolicy
+
+    env = MazeEnv({'size': 9})
+    env.seed(1)
+    obs = env.reset()
+    values, _ = get_vi_sequence(env, obs)
+    final_v = values[-1]
+    # expert: from every open cell, act toward the neighbour with higher value
+    data = []
+    for r in range(9):
+        for c in range(9):
+            if env.maze[r, c] != 0 or final_v[r, c] == -np.inf:
+                continue
+            best_a, best_val = None, final_v[r, c]
+            for a, (dr, dc) in enumerate([(-1, 0), (1, 0), (0, -1), (0, 1)]):
+                nr, nc = r + dr, c + dc
+                if 0 <= nr < 9 and 0 <= nc < 9 and env.maze[nr, nc] == 0 and final_v[nr, nc] > best_val:
+                    best_a, best_val = a, final_v[nr, nc]
+            if best_a is None:
+                continue
+            env._agent = (r, c)
+            data.append({'obs': torch.as_tensor(env._obs()).reshape(-1), 'action': torch.tensor([best_a])})
+    assert len(data) > 10
+    cfg = EasyDict(deep_merge_dicts(BehaviourCloningPolicy.default_config(), EasyDict(dict(
+        cuda=False, continuous=False,
+        model=dict(obs_shape=8 * 9 * 9, action_shape=4, encoder_hidden_size_list=[64, 64]),
+        learn=dict(batch_size=16, learning_rate=1e-3),
+    ))))
+    pol = create_policy(cfg, enable_field=['learn'])
+    out = pol._forward_learn(data[:16])
+    assert 'total_loss' in out or 'loss' in out
