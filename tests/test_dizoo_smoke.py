@@ -236,7 +236,13 @@ def test_maze_bc_with_bfs_expert():
             if best_a is None:
                 continue
             env._agent = (r, c)
-            data.append({'obs': torch.as_tensor(env._obs()).reshape(-1), 'action': torch.tensor([best_a])})
+            data.append({
+                'obs': torch.as_tensor(env._obs()).reshape(-1),
+                'next_obs': torch.as_tensor(env._obs()).reshape(-1),
+                'action': torch.tensor([best_a]),
+                'reward': torch.tensor([0.0]),
+                'done': False,
+            })
     assert len(data) > 10
     cfg = EasyDict(deep_merge_dicts(BehaviourCloningPolicy.default_config(), EasyDict(dict(
         cuda=False, continuous=False,
