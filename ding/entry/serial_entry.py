@@ -141,6 +141,9 @@ def serial_pipeline_onpolicy(
             if stop:
                 break
         new_data = collector.collect(train_iter=learner.train_iter)
+        if new_data and isinstance(new_data[0], list):
+            # episode collector with get_train_sample: flatten episodes
+            new_data = [t for episode in new_data for t in episode]
         learner.train(new_data, collector.envstep)
         if collector.envstep >= max_env_step or learner.train_iter >= max_train_iter:
             break

@@ -1,0 +1,31 @@
+"""Pendulum DDPG (reference pendulum_ddpg_config.py, stop_value -250)."""
+from ding.utils import EasyDict
+
+pendulum_ddpg_config = EasyDict(dict(
+    exp_name='pendulum_ddpg_seed0',
+    env=dict(
+        collector_env_num=8,
+        evaluator_env_num=5,
+        n_evaluator_episode=5,
+        stop_value=-250,
+        act_scale=True,
+    ),
+    policy=dict(
+        cuda=False,
+        random_collect_size=800,
+        model=dict(obs_shape=3, action_shape=1, action_space='regression', twin_critic=False),
+        learn=dict(update_per_collect=2, batch_size=128, learning_rate_actor=1e-3,
+                   learning_rate_critic=1e-3, target_theta=0.005, discount_factor=0.99,
+                   actor_update_freq=1, noise=False),
+        collect=dict(n_sample=48, unroll_len=1, noise_sigma=0.1),
+        eval=dict(evaluator=dict(eval_freq=100, )),
+        other=dict(replay_buffer=dict(replay_buffer_size=20000)),
+    ),
+))
+main_config = pendulum_ddpg_config
+pendulum_ddpg_create_config = EasyDict(dict(
+    env=dict(type='pendulum', import_names=['dizoo.classic_control.pendulum.envs.pendulum_env']),
+    env_manager=dict(type='base'),
+    policy=dict(type='ddpg'),
+))
+create_config = pendulum_ddpg_create_config

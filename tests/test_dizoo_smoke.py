@@ -92,6 +92,23 @@ SMOKE_CONFIGS = [
     ('dizoo.dmc2gym.config.dmc2gym_sac_pixel_config', 'serial'),
     ('dizoo.dmc2gym.config.dmc2gym_ppo_config', 'onpolicy'),
     ('dizoo.cliffwalking.config.cliffwalking_dqn_config', 'serial'),
+    ('dizoo.classic_control.cartpole.config.cartpole_c51_config', 'serial'),
+    ('dizoo.classic_control.cartpole.config.cartpole_qrdqn_config', 'serial'),
+    ('dizoo.classic_control.cartpole.config.cartpole_iqn_config', 'serial'),
+    ('dizoo.classic_control.cartpole.config.cartpole_fqf_config', 'serial'),
+    ('dizoo.classic_control.cartpole.config.cartpole_rainbow_config', 'serial'),
+    ('dizoo.classic_control.cartpole.config.cartpole_mdqn_config', 'serial'),
+    ('dizoo.classic_control.cartpole.config.cartpole_sqn_config', 'serial'),
+    ('dizoo.classic_control.cartpole.config.cartpole_pg_config', 'onpolicy'),
+    ('dizoo.classic_control.cartpole.config.cartpole_ppg_offpolicy_config', 'serial'),
+    ('dizoo.classic_control.cartpole.config.cartpole_acer_config', 'serial'),
+    ('dizoo.classic_control.cartpole.config.cartpole_dqn_stdim_config', 'serial'),
+    ('dizoo.classic_control.cartpole.config.cartpole_ppo_stdim_config', 'onpolicy'),
+    ('dizoo.classic_control.cartpole.config.cartpole_dqfd_config', 'dqfd'),
+    ('dizoo.classic_control.pendulum.config.pendulum_ddpg_config', 'serial'),
+    ('dizoo.classic_control.pendulum.config.pendulum_td3_config', 'serial'),
+    ('dizoo.classic_control.pendulum.config.pendulum_sac_config', 'serial'),
+    ('dizoo.classic_control.pendulum.config.pendulum_d4pg_config', 'serial'),
 ]
 
 
@@ -143,6 +160,16 @@ def _run_one(module_name: str, pipeline: str, tmp_dir: str = None):
     elif pipeline == 'ngu':
         from ding.entry import serial_pipeline_ngu
         serial_pipeline_ngu((m, c), seed=0, max_train_iter=1)
+    elif pipeline == 'dqfd':
+        # expert = a fresh DQN on the same env (1-iter smoke scale)
+        from ding.entry import serial_pipeline_dqfd, collect_demo_data
+        import copy as _copy
+        em, ec = _copy.deepcopy(m), _copy.deepcopy(c)
+        ec.policy.type = 'dqn'
+        import os as _os
+        expert_path = _os.path.join(tmp_dir or '.', 'dqfd_expert.pkl')
+        collect_demo_data((em, ec), seed=0, collect_count=32, expert_data_path=expert_path)
+        serial_pipeline_dqfd((m, c), expert_path, seed=0, max_train_iter=1)
     elif pipeline == 'offline':
         from ding.entry import serial_pipeline_offline
         serial_pipeline_offline((m, c), seed=0, max_train_iter=1)
