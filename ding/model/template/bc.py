@@ -13,6 +13,12 @@ def _enc(obs_shape, hidden_size_list, activation, norm_type):
     if isinstance(obs_shape, int) or len(obs_shape) == 1:
         return FCEncoder(squeeze(obs_shape), hidden_size_list, activation=activation, norm_type=norm_type), \
             hidden_size_list[-1]
+    if min(obs_shape[1:]) < 32:
+        # small maps (e.g. the 16x16 maze): the Atari 8/4/3 stack underflows
+        return ConvEncoder(
+            obs_shape, hidden_size_list, activation=activation, norm_type=norm_type,
+            kernel_size=[3, 3, 3], stride=[2, 2, 1]
+        ), hidden_size_list[-1]
     return ConvEncoder(obs_shape, hidden_size_list, activation=activation, norm_type=norm_type), hidden_size_list[-1]
 
 

@@ -100,7 +100,7 @@ class MBSACPolicy(SACPolicy):
             rewards = torch.cat([data['reward'].unsqueeze(0), rewards])
             aug_rewards = torch.cat([torch.zeros_like(data['reward']).unsqueeze(0), aug_rewards])
             dones = torch.cat([data['done'].unsqueeze(0), dones])
-        dones = torch.cat([torch.zeros_like(dones[0]).unsqueeze(0), dones])
+        dones = torch.cat([torch.zeros_like(dones[0]).unsqueeze(0), dones]).float()
 
         # (T+1, B) target values + entropy bonus
         target_q_values = q_evaluation(obss, actions, partial(self._critic_fn, model=self._target_model))

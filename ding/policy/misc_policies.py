@@ -263,6 +263,32 @@ class TD3VAEPolicy(TD3Policy):
             self._vae.cuda()
         self._optimizer_vae = Adam(self._vae.parameters(), lr=self._cfg.learn.learning_rate_vae)
 
+    def _forward_learn(self, data: List[Dict[str, Any]]) -> Dict[str, Any]:
+        """Phase-aware learn (serial_pipeline_td3_vae): warm_up / vae_phase
+        batches train the VAE; rl_phase batches run TD3 in the latent action
+        space (real actions encoded through the frozen VAE)."""
+        if data and isinstance(data[0], dict) and (data[0].get('warm_up') or data[0].get('vae_phase')):
+            return self.train_vae(data)
+        if data and isinstance(data[0], dict) and data[0].get('rl_phase'):
+            data = [dict(d) for d in data]
+            need_encode = [d for d in data if 'latent_action' not in d]
+            if need_encode:
+                with torch.no_grad():
+                    obs = torch.stack([torch.as_tensor(d['obs'], dtype=torch.float32) for d in need_encode])
+                    act = torch.stack(
+                        [torch.as_tensor(d['action'], dtype=torch.float32).reshape(-1) for d in need_encode]
+                    )
+                    if self._cuda:
+                        obs, act = obs.cuda(), act.cuda()
+                    latent = self._vae.encode({'obs': obs, 'action': act})['mu'].cpu()
+                for d, z in zip(need_encode, latent):
+                    d['latent_action'] = z
+            for d in data:
+                d['action'] = torch.as_tensor(d['latent_action'], dtype=torch.float32).reshape(-1)
+                for k in ('warm_up', 'rl_phase', 'vae_phase', 'latent_action'):
+                    d.pop(k, None)
+        return super()._forward_learn(data)
+
     def train_vae(self, data: List[Dict[str, Any]]) -> Dict[str, float]:
         collated = default_preprocess_learn(data, use_nstep=False)
         if self._cuda:
@@ -276,6 +302,38 @@ class TD3VAEPolicy(TD3Policy):
         losses['loss'].backward()
         self._optimizer_vae.step()
         return {'vae_loss': losses['loss'].item()}
+
+    def _forward_collect(self, data: Dict[int, Any], **kwargs) -> Dict[int, Any]:
+        """TD3 acts in the latent space; the VAE decoder maps the latent to
+        the env's real action. Both are kept in the output so learn gets
+        the latent and the env gets the real action (HyAR)."""
+        out = super()._forward_collect(data, **kwargs)
+        with torch.no_grad():
+            for env_id, o in out.items():
+                z = torch.as_tensor(o['action'], dtype=torch.float32).reshape(1, -1)
+                obs = torch.as_tensor(data[env_id], dtype=torch.float32).reshape(1, -1)
+                dec = self._vae.decode({'obs': obs, 'z': z})['reconstruction_action']
+                o['latent_action'] = o['action']
+                o['action'] = dec.reshape(-1).cpu()
+        return out
+
+    def _process_transition(self, obs, policy_output, timestep):
+        tr = super()._process_transition(obs, policy_output, timestep)
+        if 'latent_action' in policy_output:
+            tr['latent_action'] = policy_output['latent_action']
+        return tr
+
+    def _init_eval(self) -> None:
+        super()._init_eval()
+
+    def _forward_eval(self, data: Dict[int, Any]) -> Dict[int, Any]:
+        out = super()._forward_eval(data)
+        with torch.no_grad():
+            for env_id, o in out.items():
+                z = torch.as_tensor(o['action'], dtype=torch.float32).reshape(1, -1)
+                obs = torch.as_tensor(data[env_id], dtype=torch.float32).reshape(1, -1)
+                o['action'] = self._vae.decode({'obs': obs, 'z': z})['reconstruction_action'].reshape(-1).cpu()
+        return out
 
 
 # -------------------------------------------------------------- LLM prompts

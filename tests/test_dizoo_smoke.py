@@ -109,6 +109,14 @@ SMOKE_CONFIGS = [
     ('dizoo.classic_control.pendulum.config.pendulum_td3_config', 'serial'),
     ('dizoo.classic_control.pendulum.config.pendulum_sac_config', 'serial'),
     ('dizoo.classic_control.pendulum.config.pendulum_d4pg_config', 'serial'),
+    ('dizoo.classic_control.pendulum.config.pendulum_td3_vae_config', 'td3_vae'),
+    ('dizoo.classic_control.pendulum.config.pendulum_ibc_config', 'offline_gen'),
+    ('dizoo.classic_control.cartpole.config.cartpole_discrete_cql_config', 'offline_gen'),
+    ('dizoo.classic_control.pendulum.config.pendulum_mbsac_ddppo_config', 'dream'),
+    ('dizoo.classic_control.pendulum.config.pendulum_stevesac_ddppo_config', 'dream'),
+    ('dizoo.mujoco.config.hopper_bdq_config', 'serial'),
+    ('dizoo.maze.config.maze_pc_config', 'pc'),
+    ('dizoo.dmc2gym.config.cartpole_balance_dreamer_config', 'dreamer'),
 ]
 
 
@@ -160,6 +168,93 @@ def _run_one(module_name: str, pipeline: str, tmp_dir: str = None):
     elif pipeline == 'ngu':
         from ding.entry import serial_pipeline_ngu
         serial_pipeline_ngu((m, c), seed=0, max_train_iter=1)
+    elif pipeline == 'td3_vae':
+        from ding.entry import serial_pipeline_td3_vae
+        m.policy.learn.warm_up_update = 8
+        m.policy.random_collect_size = 16
+        serial_pipeline_td3_vae((m, c), seed=0, max_train_iter=1)
+    elif pipeline == 'offline_gen':
+        # synthesize a matching-shape npz for offline policies
+        import os
+        import numpy as np
+        from ding.entry import serial_pipeline_offline
+        path = os.path.join(tmp_dir or '.', 'gen.npz')
+        rng = np.random.RandomState(0)
+        obs_shape = m.policy.model.obs_shape
+        obs_dim = obs_shape if isinstance(obs_shape, int) else int(np.prod(obs_shape))
+        n = 256
+        act_shape = m.policy.model.action_shape
+        discrete = c.policy.type in ('discrete_cql', 'bc')
+        action = rng.randint(0, act_shape, n) if discrete else \
+            np.tanh(rng.randn(n, act_shape)).astype(np.float32)
+        np.savez(path, obs=rng.randn(n, obs_dim).astype(np.float32), action=action,
+                 reward=rng.randn(n).astype(np.float32), done=(rng.rand(n) < 0.02),
+                 next_obs=rng.randn(n, obs_dim).astype(np.float32))
+        m.policy.collect.data_path = path
+        serial_pipeline_offline((m, c), seed=0, max_train_iter=1)
+    elif pipeline == 'dream':
+        from ding.entry import serial_pipeline_dream
+        m.world_model.model.batch_size = 16
+        m.world_model.model.ensemble_size = 2
+        m.world_model.model.elite_size = 1
+        if m.policy.learn.get('ensemble_size') is not None:
+            m.policy.learn.ensemble_size = 2
+        m.world_model.model.hidden_size = 32
+        m.world_model.train_freq = 4
+        m.world_model.eval_freq = 100
+        m.policy.random_collect_size = 24
+        serial_pipeline_dream((m, c), seed=0, max_train_iter=1)
+    elif pipeline == 'dreamer':
+        from ding.entry import serial_pipeline_dreamer
+        m.policy.random_collect_size = 24
+        m.policy.imag_horizon = 4
+        m.policy.model.update(dict(dyn_stoch=8, dyn_deter=32, dyn_discrete=8, units=32,
+                                   actor_layers=1, value_layers=1))
+        m.policy.learn.update(dict(batch_size=4, batch_length=6))
+        m.policy.collect.n_sample = 16
+        m.world_model.pretrain = 1
+        m.world_model.model.update(dict(dyn_stoch=8, dyn_deter=32, dyn_hidden=32, dyn_discrete=8,
+                                        units=32, reward_layers=1, discount_layers=1,
+                                        image_dec_layers=1, batch_size=4, batch_length=6,
+                                        encoder_hidden_size_list=[32, 32]))
+        serial_pipeline_dreamer((m, c), seed=0, max_train_iter=1)
+    elif pipeline == 'pc':
+        # BFS expert batches from the maze env (reference maze PC data path)
+        import numpy as np
+        import torch
+        from ding.entry import serial_pipeline_pc
+        from dizoo.maze.envs.maze_env import MazeEnv
+        from ding.utils.misc_helpers import get_vi_sequence
+
+        def dataset_fn():
+            env = MazeEnv({'size': m.env.size})
+            env.seed(0)
+            obs = env.reset()
+            values, _ = get_vi_sequence(env, obs)
+            final_v = values[-1]
+            n = m.env.size
+            batch = []
+            for r in range(n):
+                for c2 in range(n):
+                    if env.maze[r, c2] != 0 or final_v[r, c2] == -np.inf:
+                        continue
+                    best_a, best_val = None, final_v[r, c2]
+                    for a, (dr, dc) in enumerate([(-1, 0), (1, 0), (0, -1), (0, 1)]):
+                        nr, nc = r + dr, c2 + dc
+                        if 0 <= nr < n and 0 <= nc < n and env.maze[nr, nc] == 0 \
+                                and final_v[nr, nc] > best_val:
+                            best_a, best_val = a, final_v[nr, nc]
+                    if best_a is None:
+                        continue
+                    env._agent = (r, c2)
+                    batch.append({'obs': torch.as_tensor(env._obs()), 'action': torch.tensor([best_a]),
+                                  'next_obs': torch.as_tensor(env._obs()),
+                                  'reward': torch.tensor([0.0]), 'done': False})
+            yield batch[:32]
+
+        m.policy.learn.dataset_fn = dataset_fn
+        m.policy.learn.train_epoch = 1
+        serial_pipeline_pc((m, c), seed=0, max_iter=2)
     elif pipeline == 'dqfd':
         # expert = a fresh DQN on the same env (1-iter smoke scale)
         from ding.entry import serial_pipeline_dqfd, collect_demo_data
