@@ -530,6 +530,7 @@ class ILPolicy(Policy):
         self._collect_model.eval()
         self._collect_model.reset()
         self._gamma = self._cfg.collect.discount_factor
+        self._unroll_len = self._cfg.collect.unroll_len
 
     def _forward_collect(self, data: dict, **kwargs) -> dict:
         data_id = list(data.keys())

@@ -60,6 +60,8 @@ def to_tensor(item: Any, dtype: Optional[torch.dtype] = None, ignore_keys: list 
             return [] if isinstance(item, list) else ()
         if hasattr(item, "_fields"):  # namedtuple
             return type(item)(*[to_tensor(v, dtype) for v in item])
+        if isinstance(item[0], str):
+            return item  # text payloads (prompt policies) pass through
         if np.isscalar(item[0]):
             return torch.as_tensor(item, dtype=dtype if dtype is not None else torch.float32)
         return type(item)(to_tensor(v, dtype, ignore_keys, transform_scalar) for v in item)

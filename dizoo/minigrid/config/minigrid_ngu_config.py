@@ -64,6 +64,6 @@ main_config = minigrid_ngu_config
 minigrid_ngu_create_config = EasyDict(dict(
     env=dict(type='minigrid', import_names=['dizoo.minigrid.envs.minigrid_env']),
     env_manager=dict(type='subprocess'),
-    policy=dict(type='r2d2'),
+    policy=dict(type='ngu'),
 ))
 create_config = minigrid_ngu_create_config

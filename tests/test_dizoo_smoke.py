@@ -117,6 +117,12 @@ SMOKE_CONFIGS = [
     ('dizoo.mujoco.config.hopper_bdq_config', 'serial'),
     ('dizoo.maze.config.maze_pc_config', 'pc'),
     ('dizoo.dmc2gym.config.cartpole_balance_dreamer_config', 'dreamer'),
+    ('dizoo.tabmwp.config.tabmwp_prompt_pg_config', 'onpolicy'),
+    ('dizoo.classic_control.cartpole.config.cartpole_ppo_pg_config', 'onpolicy'),
+    ('dizoo.classic_control.cartpole.config.cartpole_r2d2_gtrxl_config', 'serial'),
+    ('dizoo.classic_control.cartpole.config.cartpole_r2d3_config', 'r2d3'),
+    ('dizoo.classic_control.pendulum.config.pendulum_sqil_sac_config', 'sqil'),
+    ('dizoo.league_demo.league_demo_il_config', 'serial'),
 ]
 
 
@@ -168,6 +174,19 @@ def _run_one(module_name: str, pipeline: str, tmp_dir: str = None):
     elif pipeline == 'ngu':
         from ding.entry import serial_pipeline_ngu
         serial_pipeline_ngu((m, c), seed=0, max_train_iter=1)
+    elif pipeline == 'r2d3':
+        from ding.entry import serial_pipeline_r2d3
+        import copy as _copy
+        em, ec = _copy.deepcopy(m), _copy.deepcopy(c)
+        ec.policy.type = 'r2d2'
+        serial_pipeline_r2d3((m, c), (em, ec), seed=0, max_train_iter=1)
+    elif pipeline == 'sqil':
+        from ding.entry import serial_pipeline_sqil
+        import copy as _copy
+        em, ec = _copy.deepcopy(m), _copy.deepcopy(c)
+        ec.policy.type = 'sac'
+        em.policy.random_collect_size = 16
+        serial_pipeline_sqil((m, c), (em, ec), seed=0, max_train_iter=1)
     elif pipeline == 'td3_vae':
         from ding.entry import serial_pipeline_td3_vae
         m.policy.learn.warm_up_update = 8
