@@ -29,6 +29,7 @@ class ParticleSpreadEnv(BaseEnv):
         self.landmark_num = cfg.get('landmark_num', self.agent_num)
         self.max_step = cfg.get('max_step', 25)
         self.collide_penalty = cfg.get('collide_penalty', 1.0)
+        self.continuous_actions = cfg.get('continuous_actions', False)
         self.dt = 0.1
         self.damping = 0.25
         self.accel = 5.0
@@ -75,8 +76,11 @@ class ParticleSpreadEnv(BaseEnv):
     _THRUST = np.array([[0, 0], [-1, 0], [1, 0], [0, -1], [0, 1]], dtype=np.float32)
 
     def step(self, action: Any) -> BaseEnvTimestep:
-        action = np.asarray(action).reshape(-1).astype(np.int64)
-        force = self._THRUST[action] * self.accel
+        if self.continuous_actions:
+            force = np.clip(np.asarray(action, dtype=np.float32).reshape(self.agent_num, 2), -1, 1) * self.accel
+        else:
+            action = np.asarray(action).reshape(-1).astype(np.int64)
+            force = self._THRUST[action] * self.accel
         self._vel = self._vel * (1 - self.damping) + force * self.dt
         self._pos = self._pos + self._vel * self.dt
         # coverage reward: each landmark scored by its nearest agent
@@ -97,6 +101,8 @@ class ParticleSpreadEnv(BaseEnv):
         pass
 
     def random_action(self) -> np.ndarray:
+        if self.continuous_actions:
+            return self._rng.uniform(-1, 1, size=(self.agent_num, 2)).astype(np.float32)
         return self._rng.randint(0, 5, size=(self.agent_num, ))
 
     @property

@@ -69,6 +69,7 @@ SMOKE_CONFIGS = [
     ('dizoo.petting_zoo.config.ptz_simple_spread_mappo_config', 'onpolicy'),
     ('dizoo.petting_zoo.config.ptz_simple_spread_happo_config', 'onpolicy'),
     ('dizoo.petting_zoo.config.ptz_simple_spread_madqn_config', 'serial'),
+    ('dizoo.petting_zoo.config.ptz_simple_spread_atoc_config', 'serial'),
     ('dizoo.procgen.config.coinrun_dqn_config', 'serial'),
     ('dizoo.procgen.config.coinrun_ppo_config', 'onpolicy'),
     ('dizoo.procgen.config.coinrun_ppg_config', 'onpolicy_ppg'),
