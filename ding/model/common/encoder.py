@@ -115,6 +115,11 @@ class ConvEncoder(nn.Module):
     ):
         super().__init__()
         self.obs_shape = obs_shape
+        if kernel_size == [8, 4, 3] and stride == [4, 2, 1] and min(obs_shape[1:]) < 36:
+            # the Atari stack underflows on small maps (maze/sokoban/procgen
+            # crops): fall back to a 3x3 stack that fits >= 5x5 inputs
+            kernel_size = [3, 3, 3]
+            stride = [2, 2, 1] if min(obs_shape[1:]) >= 12 else [1, 1, 1]
         if padding is None:
             padding = [0] * len(kernel_size)
         if fast_im2col is None:

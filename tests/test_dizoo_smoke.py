@@ -124,6 +124,10 @@ SMOKE_CONFIGS = [
     ('dizoo.classic_control.cartpole.config.cartpole_r2d3_config', 'r2d3'),
     ('dizoo.classic_control.pendulum.config.pendulum_sqil_sac_config', 'sqil'),
     ('dizoo.league_demo.league_demo_il_config', 'serial'),
+    ('dizoo.bsuite.config.deep_sea_dqn_config', 'serial'),
+    ('dizoo.bsuite.config.memory_len_r2d2_config', 'serial'),
+    ('dizoo.sokoban.config.sokoban_dqn_config', 'serial'),
+    ('dizoo.ising_env.config.ising_mf_qmix_config', 'serial'),
 ]
 
 
