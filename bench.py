@@ -130,7 +130,7 @@ def ppo_step(policy, device: str, n_sample: int):
     return n_sample
 
 
-def build_impala(device: str, multi_gpu: bool):
+def build_impala(device: str, multi_gpu: bool, bf16: bool = False):
     """IMPALA workload through the real IMPALAPolicy (reference
     spaceinvaders_impala_config.py:17-45: unroll_len=32, batch=128,
     encoder [128,128,256])."""
@@ -158,6 +158,7 @@ def build_impala(device: str, multi_gpu: bool):
             entropy_weight=0.01,
             discount_factor=0.99,
             lambda_=0.95,
+            bf16=bf16,
             # hipGraph-capture the learn step (single-process only)
             cuda_graph=os.environ.get('DING_IMPALA_GRAPH', '1') not in ('0', 'false'),
         ),
@@ -230,7 +231,7 @@ def main():
             'parallelism': f'dp{world_size}', 'minibatch': 320, 'epoch_per_collect': 10,
         }
     else:
-        policy = build_impala(device, multi_gpu=distributed)
+        policy = build_impala(device, multi_gpu=distributed, bf16=args.dtype == 'bf16')
         step_fn = lambda: impala_step(policy, device)
         config = {
             'model': 'spaceinvaders_impala(conv[128,128,256] 4x84x84)', 'global_batch': 128 * 32 * world_size,

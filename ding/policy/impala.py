@@ -74,6 +74,7 @@ class IMPALAPolicy(Policy):
         # hipGraph capture of the whole learn step (fwd + v-trace + bwd):
         # static [T, B] shapes make IMPALA a one-graph-per-step workload
         self._cuda_graph = self._cfg.learn.get('cuda_graph', False) and not self._cfg.multi_gpu
+        self._bf16 = self._cfg.learn.get('bf16', False)
         self._graphed_step = None
         self._learn_model.reset()
 
@@ -99,7 +100,14 @@ class IMPALAPolicy(Policy):
         sync-free (returns 0-dim GPU tensors) so it can be hipGraph-captured."""
         T, B = data['done'].shape[:2]
         obs_flat = data['obs_plus_1'].reshape(-1, *data['obs_plus_1'].shape[2:])
-        output = self._learn_model.forward(obs_flat, mode='compute_actor_critic')
+        if self._bf16:
+            # opt-in bf16 lane: conv/GEMM fwd+bwd in bf16, fp32 master
+            # weights, v-trace loss math in fp32
+            with torch.autocast(obs_flat.device.type, dtype=torch.bfloat16):
+                output = self._learn_model.forward(obs_flat, mode='compute_actor_critic')
+            output = {k: v.float() for k, v in output.items()}
+        else:
+            output = self._learn_model.forward(obs_flat, mode='compute_actor_critic')
         target_logit = output['logit'].reshape(T + 1, B, -1)[:-1]
         value = output['value'].reshape(T + 1, B)
         rewards = data['reward']
