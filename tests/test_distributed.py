@@ -385,3 +385,51 @@ def _body_trajectory_shipper(rank, world):
 def test_trajectory_shipper_two_ranks():
     res = _run_dist('_body_trajectory_shipper')
     assert abs(res[0]['sent_obs_sum'] - res[1]['recv_obs_sum']) < 1e-5
+
+
+def _body_gpu_exchanger_actor_learner(rank, world):
+    """Actor rank 0 collects fake IMPALA unrolls and ships them through the
+    gpu exchanger middleware; learner rank 1 trains IMPALAPolicy on the
+    received collated batch (the same-node fast path, gloo here / RCCL on
+    a real node)."""
+    from ding.framework.middleware import gpu_trajectory_sender, gpu_trajectory_receiver
+    T, B, N = 4, 3, 2
+
+    class Ctx:
+        pass
+
+    if rank == 0:
+        send = gpu_trajectory_sender(dst=1, collate=False)
+        ctx = Ctx()
+        ctx.env_step = 12
+        ctx.trajectories = None
+        ctx.train_data = {
+            'obs_plus_1': torch.randn(T + 1, B, 4),
+            'logit': torch.randn(T, B, N),
+            'action': torch.randint(0, N, (T, B)),
+            'reward': torch.randn(T, B),
+            'done': torch.zeros(T, B),
+        }
+        send(ctx)
+        return {'sent': float(ctx.train_data['obs_plus_1'].sum())}
+    else:
+        from ding.policy import IMPALAPolicy
+        from ding.utils import EasyDict, deep_merge_dicts
+        recv = gpu_trajectory_receiver(src=0)
+        ctx = Ctx()
+        ctx.env_step = 0
+        recv(ctx)
+        assert ctx.env_step == 12
+        cfg = EasyDict(deep_merge_dicts(IMPALAPolicy.default_config(), EasyDict(dict(
+            cuda=False, model=dict(obs_shape=4, action_shape=N, encoder_hidden_size_list=[16, 16]),
+            learn=dict(batch_size=B),
+        ))))
+        pol = IMPALAPolicy(cfg, enable_field=['learn'])
+        out = pol._forward_learn(ctx.train_data)
+        assert 'total_loss' in out
+        return {'recv': float(ctx.train_data['obs_plus_1'].sum())}
+
+
+def test_gpu_exchanger_actor_learner():
+    res = _run_dist('_body_gpu_exchanger_actor_learner')
+    assert abs(res[0]['sent'] - res[1]['recv']) < 1e-4
