@@ -128,6 +128,8 @@ SMOKE_CONFIGS = [
     ('dizoo.bsuite.config.memory_len_r2d2_config', 'serial'),
     ('dizoo.sokoban.config.sokoban_dqn_config', 'serial'),
     ('dizoo.ising_env.config.ising_mf_qmix_config', 'serial'),
+    ('dizoo.gym_anytrading.config.stocks_dqn_config', 'serial'),
+    ('dizoo.beergame.config.beergame_onppo_config', 'onpolicy'),
 ]
 
 
@@ -398,3 +400,27 @@ def test_maze_bc_with_bfs_expert():
     pol = create_policy(cfg, enable_field=['learn'])
     out = pol._forward_learn(data[:16])
     assert 'total_loss' in out or 'loss' in out
+
+
+def test_image_classification_supervised():
+    """Supervised lane: CE training + metric evaluation accuracy improves on
+    the synthetic digits (reference dizoo/image_classification)."""
+    import torch
+    from torch.utils.data import DataLoader
+    from dizoo.image_classification.policy import ImageClassificationPolicy
+    from dizoo.image_classification.data.dataset import SyntheticDigits
+    from ding.utils import EasyDict, deep_merge_dicts
+    cfg = EasyDict(deep_merge_dicts(ImageClassificationPolicy.default_config(), EasyDict(dict(
+        cuda=False,
+        model=dict(obs_shape=[1, 28, 28], action_shape=10, encoder_hidden_size_list=[16, 16, 32]),
+        learn=dict(batch_size=64, learning_rate=1e-3),
+    ))))
+    pol = ImageClassificationPolicy(cfg, enable_field=['learn', 'eval'])
+    ds = SyntheticDigits(n=512, seed=3)
+    loader = DataLoader(ds, batch_size=64, shuffle=True, collate_fn=lambda b: b)
+    accs = []
+    for epoch in range(8):
+        for batch in loader:
+            out = pol._forward_learn(batch)
+        accs.append(out['acc'])
+    assert accs[-1] > 0.5, f"classifier failed to learn: {accs}"
