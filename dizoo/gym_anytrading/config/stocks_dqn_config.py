@@ -15,7 +15,7 @@ stocks_dqn_config = EasyDict(dict(
     ),
     policy=dict(
         cuda=True,
-        model=dict(obs_shape=[20, 2], action_shape=2, encoder_hidden_size_list=[128, 128, 64]),
+        model=dict(obs_shape=40, action_shape=2, encoder_hidden_size_list=[128, 128, 64]),
         nstep=3,
         discount_factor=0.99,
         learn=dict(update_per_collect=10, batch_size=64, learning_rate=1e-4, target_update_freq=500),

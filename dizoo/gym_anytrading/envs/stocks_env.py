@@ -22,7 +22,7 @@ class StocksEnv(BaseEnv):
         self.window = int(self._cfg.get('window_size', 20))
         self.horizon = int(self._cfg.get('eps_length', 200))
         self.commission = float(self._cfg.get('commission', 0.001))
-        self._observation_space = Box(-np.inf, np.inf, (self.window, 2))
+        self._observation_space = Box(-np.inf, np.inf, (self.window * 2, ))
         self._action_space = Discrete(2)
         self._reward_space = Box(-1.0, 1.0, (1, ))
         self._rng = np.random.RandomState()
@@ -49,7 +49,7 @@ class StocksEnv(BaseEnv):
     def _obs(self) -> np.ndarray:
         diffs = np.diff(self.prices[self.t - self.window:self.t + 1]) / self.prices[self.t - self.window:self.t]
         pos = np.full(self.window, float(self.position), dtype=np.float32)
-        return np.stack([diffs.astype(np.float32) * 100.0, pos], axis=-1)
+        return np.stack([diffs.astype(np.float32) * 100.0, pos], axis=-1).reshape(-1)
 
     def step(self, action: Any) -> BaseEnvTimestep:
         if hasattr(action, 'reshape'):
