@@ -130,6 +130,7 @@ SMOKE_CONFIGS = [
     ('dizoo.ising_env.config.ising_mf_qmix_config', 'serial'),
     ('dizoo.gym_anytrading.config.stocks_dqn_config', 'serial'),
     ('dizoo.beergame.config.beergame_onppo_config', 'onpolicy'),
+    ('dizoo.competitive_rl.config.cpong_dqn_config', 'serial'),
 ]
 
 
