@@ -149,6 +149,12 @@ def ppo_policy_error_continuous(
     dist_old = _gaussian_dist(mu_sigma_old)
     logp_new = dist_new.log_prob(action)
     logp_old = dist_old.log_prob(action)
+    # multi-agent continuous: log_prob carries a trailing agent dim that
+    # adv/weight ([B]) lack — broadcast them across agents
+    while adv.dim() < logp_new.dim():
+        adv = adv.unsqueeze(-1)
+    while weight.dim() < logp_new.dim():
+        weight = weight.unsqueeze(-1)
     entropy_loss = (dist_new.entropy() * weight).mean()
     ratio = torch.exp(logp_new - logp_old)
     surr1 = ratio * adv

@@ -131,6 +131,7 @@ SMOKE_CONFIGS = [
     ('dizoo.gym_anytrading.config.stocks_dqn_config', 'serial'),
     ('dizoo.beergame.config.beergame_onppo_config', 'onpolicy'),
     ('dizoo.competitive_rl.config.cpong_dqn_config', 'serial'),
+    ('dizoo.multiagent_mujoco.config.mamujoco_mappo_config', 'onpolicy'),
 ]
 
 
