@@ -132,6 +132,11 @@ SMOKE_CONFIGS = [
     ('dizoo.beergame.config.beergame_onppo_config', 'onpolicy'),
     ('dizoo.competitive_rl.config.cpong_dqn_config', 'serial'),
     ('dizoo.multiagent_mujoco.config.mamujoco_mappo_config', 'onpolicy'),
+    ('dizoo.rocket.config.rocket_hover_ppo_config', 'onpolicy'),
+    ('dizoo.gym_soccer.config.gym_soccer_pdqn_config', 'serial'),
+    ('dizoo.pybullet.config.hopper_bullet_sac_config', 'serial'),
+    ('dizoo.gym_pybullet_drones.config.drone_hover_td3_config', 'serial'),
+    ('dizoo.overcooked.config.overcooked_qmix_config', 'serial'),
 ]
 
 
