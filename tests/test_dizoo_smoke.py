@@ -137,6 +137,7 @@ SMOKE_CONFIGS = [
     ('dizoo.pybullet.config.hopper_bullet_sac_config', 'serial'),
     ('dizoo.gym_pybullet_drones.config.drone_hover_td3_config', 'serial'),
     ('dizoo.overcooked.config.overcooked_qmix_config', 'serial'),
+    ('dizoo.atari.config.serial.pong_dqn_config', 'serial'),
 ]
 
 
