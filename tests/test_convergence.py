@@ -115,3 +115,45 @@ def test_cartpole_dqn_bf16_learn_step():
     assert all(p.dtype == torch.float32 for p in pol._model.parameters()), "master weights stay fp32"
     import math
     assert math.isfinite(out['total_loss'])
+
+
+def test_cliffwalking_dqn_learns_optimal_path():
+    """Tabular gate: DQN reaches near-optimal (-13) cliffwalking returns."""
+    from ding.entry import serial_pipeline, eval as eval_entry
+    import copy
+    from dizoo.cliffwalking.config.cliffwalking_dqn_config import main_config, create_config
+    main, create = copy.deepcopy(main_config), copy.deepcopy(create_config)
+    main.exp_name = 'exp/conv_cliff_dqn'
+    main.env.collector_env_num = 4
+    main.env.evaluator_env_num = 2
+    main.env.n_evaluator_episode = 2
+    main.env.stop_value = -15
+    main.policy.other.eps = dict(type='exp', start=0.95, end=0.05, decay=5000)
+    main.policy.learn.learning_rate = 5e-4
+    main.policy.discount_factor = 0.99
+    create.env_manager.type = 'base'
+    serial_pipeline((main, create), seed=0, max_env_step=100000)
+    import glob
+    ckpts = glob.glob(f'{main.exp_name}*/ckpt/ckpt_best.pth.tar')
+    assert ckpts
+    value = eval_entry((main, create), seed=0, load_path=sorted(ckpts)[-1])
+    assert value >= -30, f"cliffwalking best-ckpt eval {value} < -30"
+
+
+def test_frozen_lake_dqn_converges():
+    """Tabular gate: non-slippery FrozenLake solved (success rate ~1)."""
+    from ding.entry import serial_pipeline, eval as eval_entry
+    import copy
+    from dizoo.frozen_lake.config.frozen_lake_dqn_config import main_config, create_config
+    main, create = copy.deepcopy(main_config), copy.deepcopy(create_config)
+    main.exp_name = 'exp/conv_frozen_dqn'
+    main.env.collector_env_num = 4
+    main.env.evaluator_env_num = 2
+    main.env.n_evaluator_episode = 4
+    create.env_manager.type = 'base'
+    serial_pipeline((main, create), seed=0, max_env_step=30000)
+    import glob
+    ckpts = glob.glob(f'{main.exp_name}*/ckpt/ckpt_best.pth.tar')
+    assert ckpts
+    value = eval_entry((main, create), seed=0, load_path=sorted(ckpts)[-1])
+    assert value >= 0.75, f"frozen_lake best-ckpt eval {value} < 0.75"
