@@ -61,7 +61,8 @@ class IMPALAPolicy(Policy):
         else:
             self._optimizer = Adam(
                 self._model.parameters(), lr=self._cfg.learn.learning_rate,
-                grad_clip_type=self._cfg.learn.grad_clip_type, clip_value=self._cfg.learn.clip_value
+                grad_clip_type=self._cfg.learn.grad_clip_type, clip_value=self._cfg.learn.clip_value,
+                flatten_grads=self._cfg.learn.get('flatten_grads', self._cuda and not self._cfg.multi_gpu),
             )
         self._learn_model = model_wrap(self._model, wrapper_name='base')
         self._value_weight = self._cfg.learn.value_weight

@@ -91,6 +91,10 @@ class PPOPolicy(Policy):
             grad_clip_type=self._cfg.learn.grad_clip_type,
             clip_value=self._cfg.learn.grad_clip_value,
             capturable=_capturable,
+            # flat grad views: 1-kernel clip/zero_grad (rocprof: per-param
+            # reduce/fill storm ~25 ms/step); DDP keeps per-param grads for
+            # the bucketed reducer's own packing
+            flatten_grads=self._cfg.learn.get('flatten_grads', self._cuda and not self._cfg.multi_gpu),
         )
         self._optimizer_capturable = _capturable
         self._learn_model = model_wrap(self._model, wrapper_name='base')

@@ -77,6 +77,7 @@ class Adam(torch.optim.Adam):
         ignore_norm_type: float = 2.0,
         ignore_momentum_timestep: int = 100,
         capturable: bool = False,
+        flatten_grads: bool = False,
     ):
         self._grad_clip_type = grad_clip_type
         self._clip_value = clip_value
@@ -111,6 +112,20 @@ class Adam(torch.optim.Adam):
             for group in self.param_groups:
                 group.setdefault('step_count', 0)
                 group['grad_norm_ema'] = None
+        # flat gradient buffer: every param.grad becomes a view into ONE
+        # contiguous tensor, so grad-clip is 1 norm + 1 scale kernel and
+        # zero_grad is 1 fill — rocprof showed the per-param reduce/fill
+        # storm at ~25 ms/step on the PPO bench. Views have stable addresses,
+        # which also makes backward hipGraph-capture friendly.
+        self._flat_grad_buf = None
+        if flatten_grads and leaves and all(p.is_cuda for p in leaves):
+            total = sum(p.numel() for p in leaves)
+            buf = torch.zeros(total, device=leaves[0].device, dtype=leaves[0].dtype)
+            off = 0
+            for p in leaves:
+                p.grad = buf[off:off + p.numel()].view_as(p)
+                off += p.numel()
+            self._flat_grad_buf = buf
 
     def _params(self) -> List[torch.Tensor]:
         return [p for group in self.param_groups for p in group['params']]
@@ -120,6 +135,12 @@ class Adam(torch.optim.Adam):
         if t is None:
             return
         params = self._params()
+        if t == 'clip_norm' and self._flat_grad_buf is not None:
+            buf = self._flat_grad_buf
+            total_norm = torch.linalg.vector_norm(buf, self._clip_norm_type)
+            scale = (self._clip_value / (total_norm + 1e-6)).clamp(max=1.0)
+            buf.mul_(scale)
+            return
         if t == 'clip_value':
             nn.utils.clip_grad_value_(params, self._clip_value)
         elif t == 'clip_norm':
@@ -161,6 +182,13 @@ class Adam(torch.optim.Adam):
     def step(self, closure=None):
         self._apply_clip()
         return super().step(closure)
+
+    def zero_grad(self, set_to_none: bool = True):
+        if self._flat_grad_buf is not None:
+            # keep the grad views alive: one fill over the flat buffer
+            self._flat_grad_buf.zero_()
+            return
+        return super().zero_grad(set_to_none)
 
     def get_grad(self) -> float:
         return calculate_grad_norm_without_bias_two_norm(_FakeModel(self._params()))

@@ -573,3 +573,30 @@ def test_conv_wrw_timing(ext):
     # informational A/B: MIOpen's tuned wrw wins these shapes (hence the
     # kernel is opt-in, DING_NATIVE_WRW=1); only sanity-bound the gap
     assert sum(r[4] for r in results) <= sum(r[5] for r in results) * 20
+
+
+def test_flat_grad_clip_matches_reference(ext):
+    """Flat-buffer clip (1 norm + 1 scale) equals nn.utils.clip_grad_norm_."""
+    import copy
+    from ding.torch_utils import Adam
+    torch.manual_seed(0)
+    m1 = torch.nn.Sequential(torch.nn.Linear(16, 32), torch.nn.ReLU(), torch.nn.Linear(32, 4)).cuda()
+    m2 = copy.deepcopy(m1)
+    o1 = Adam(m1.parameters(), lr=1e-2, grad_clip_type='clip_norm', clip_value=0.1, flatten_grads=True)
+    o2 = Adam(m2.parameters(), lr=1e-2, grad_clip_type='clip_norm', clip_value=0.1, flatten_grads=False)
+    assert o1._flat_grad_buf is not None
+    for it in range(3):
+        x = torch.randn(8, 16, device='cuda')
+        o1.zero_grad()
+        o2.zero_grad()
+        m1(x).pow(2).mean().backward()
+        m2(x).pow(2).mean().backward()
+        o1.step()
+        o2.step()
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        assert torch.allclose(p1, p2, atol=1e-5), (p1 - p2).abs().max()
+    # grads must still be views of the flat buffer after training steps
+    base = o1._flat_grad_buf.data_ptr()
+    end = base + o1._flat_grad_buf.numel() * o1._flat_grad_buf.element_size()
+    for p in m1.parameters():
+        assert base <= p.grad.data_ptr() < end
