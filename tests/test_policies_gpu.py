@@ -193,32 +193,41 @@ def test_atari_lite_ppo_improves_on_gpu():
     assert final > baseline + 0.5, f'no improvement: baseline {baseline}, final {final}'
 
 
-def test_impala_bf16_learn_decreases_loss():
-    """IMPALA bf16 lane on GPU: losses stay finite and the policy updates
-    in bf16 compute with fp32 master weights (the 764k samples/s lane)."""
+def test_impala_bf16_matches_fp32_lane():
+    """IMPALA bf16 lane on GPU: same batch, same weights — the bf16 forward
+    produces losses matching the fp32 lane within bf16 tolerance, grads
+    update, and master weights stay fp32 (the 764k samples/s lane)."""
+    import copy
     import torch
     from ding.policy import IMPALAPolicy
     from ding.utils import EasyDict, deep_merge_dicts
     T, B, N = 8, 16, 6
-    cfg = EasyDict(deep_merge_dicts(IMPALAPolicy.default_config(), EasyDict(dict(
-        cuda=True,
-        model=dict(obs_shape=[4, 84, 84], action_shape=N, encoder_hidden_size_list=[32, 32, 64]),
-        learn=dict(batch_size=B, bf16=True, cuda_graph=False),
-    ))))
-    pol = IMPALAPolicy(cfg, enable_field=['learn'])
-    assert all(p.dtype == torch.float32 for p in pol._model.parameters())
+
+    def make(bf16):
+        cfg = EasyDict(deep_merge_dicts(IMPALAPolicy.default_config(), EasyDict(dict(
+            cuda=True,
+            model=dict(obs_shape=[4, 84, 84], action_shape=N, encoder_hidden_size_list=[32, 32, 64]),
+            learn=dict(batch_size=B, bf16=bf16, cuda_graph=False),
+        ))))
+        torch.manual_seed(7)
+        return IMPALAPolicy(cfg, enable_field=['learn'])
+
+    p16, p32 = make(True), make(False)
+    p16._model.load_state_dict(p32._model.state_dict())
     torch.manual_seed(0)
-    obs_plus_1 = torch.rand(T + 1, B, 4, 84, 84, device='cuda')
     behaviour = torch.randn(T, B, N, device='cuda')
     batch = {
-        'obs_plus_1': obs_plus_1,
+        'obs_plus_1': torch.rand(T + 1, B, 4, 84, 84, device='cuda'),
         'logit': behaviour,
         'action': behaviour.argmax(-1),
         'reward': torch.randn(T, B, device='cuda'),
         'done': torch.zeros(T, B, device='cuda'),
     }
-    losses = [pol._forward_learn(dict(batch))['total_loss'] for _ in range(8)]
-    assert all(abs(l) < 1e6 for l in losses)
-    # training on a FIXED batch must reduce the loss (grads flow through bf16)
-    assert losses[-1] < losses[0], losses
-    assert all(p.dtype == torch.float32 for p in pol._model.parameters()), "master weights stay fp32"
+    before = [p.clone() for p in p16._model.parameters()]
+    out16 = p16._forward_learn(dict(batch))
+    out32 = p32._forward_learn(dict(batch))
+    for k in ('total_loss', 'policy_loss', 'value_loss'):
+        assert abs(out16[k] - out32[k]) < 0.05 + 0.1 * abs(out32[k]), (k, out16[k], out32[k])
+    changed = any(not torch.allclose(a, b) for a, b in zip(before, p16._model.parameters()))
+    assert changed, "bf16 lane did not update parameters"
+    assert all(p.dtype == torch.float32 for p in p16._model.parameters()), "master weights stay fp32"
