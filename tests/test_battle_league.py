@@ -166,3 +166,41 @@ def test_battle_interaction_evaluator():
     stop, info = ev.eval(None, train_iter=1)
     assert not stop and len(info['eval_episode_return']) >= 3
     ev.close()
+
+
+def test_pfsp_weightings():
+    """PFSP math: harder opponents get more weight; degenerate cases uniform."""
+    from ding.league.algorithm import pfsp
+    wr = np.array([0.1, 0.5, 0.9])
+    for weighting in ('variance', 'linear', 'linear_capped', 'squared'):
+        p = pfsp(wr, weighting)
+        assert abs(p.sum() - 1.0) < 1e-9
+    sq = pfsp(wr, 'squared')
+    assert sq[0] > sq[1] > sq[2], "squared PFSP must prefer opponents we lose to"
+    var = pfsp(wr, 'variance')
+    assert var[1] == max(var), "variance PFSP must prefer 50% opponents"
+    # all-beaten: uniform fallback
+    u = pfsp(np.array([1.0, 1.0]), 'squared')
+    assert np.allclose(u, [0.5, 0.5])
+
+
+def test_elo_rating_math():
+    from ding.league.metric import EloCalculator
+    a, b = EloCalculator.get_new_rating(1200, 1200, 1)
+    assert a > 1200 > b and abs((a - 1200) + (b - 1200)) < 1e-6, "zero-sum update"
+    # beating a much stronger player moves more points
+    a2, _ = EloCalculator.get_new_rating(1200, 1600, 1)
+    assert (a2 - 1200) > (a - 1200)
+    # draws move ratings toward each other
+    a3, b3 = EloCalculator.get_new_rating(1200, 1600, 0)
+    assert a3 > 1200 and b3 < 1600
+
+
+def test_trueskill_rating_math():
+    from ding.league.metric import PlayerRating, TrueSkillCalculator
+    a, b = PlayerRating(), PlayerRating()
+    a2, b2 = TrueSkillCalculator.get_new_rating(a, b, 1)
+    assert a2.mu > a.mu and b2.mu < b.mu
+    assert a2.sigma < a.sigma, "uncertainty must shrink after a match"
+    # exposure = mu - 3 sigma grows for the winner
+    assert a2.exposure > a.exposure
