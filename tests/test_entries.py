@@ -178,3 +178,40 @@ def test_plr_pipeline():
     ))
     main.level_replay = EasyDict(dict(strategy='policy_entropy', num_seeds=8))
     serial_pipeline_plr((main, create), seed=0, max_train_iter=2)
+
+
+def test_ding_cli_serial_subprocess(tmp_path):
+    """`ding -m serial -c cfg.py -s 0` end-to-end through the console entry."""
+    import subprocess
+    import sys
+    import textwrap
+    cfg = tmp_path / 'cli_cfg.py'
+    cfg.write_text(textwrap.dedent(f"""
+        from ding.utils import EasyDict
+        main_config = EasyDict(dict(
+            exp_name='{(tmp_path / "cli_exp").as_posix()}',
+            env=dict(collector_env_num=2, evaluator_env_num=1, n_evaluator_episode=1, stop_value=195,
+                     max_step=30),
+            policy=dict(
+                cuda=False,
+                model=dict(obs_shape=4, action_shape=2, encoder_hidden_size_list=[16, 16]),
+                nstep=1, discount_factor=0.97,
+                learn=dict(update_per_collect=1, batch_size=8, learning_rate=1e-3),
+                collect=dict(n_sample=16),
+                eval=dict(evaluator=dict(eval_freq=int(1e6))),
+                other=dict(eps=dict(type='exp', start=0.95, end=0.1, decay=10000),
+                           replay_buffer=dict(replay_buffer_size=1000)),
+            ),
+        ))
+        create_config = EasyDict(dict(
+            env=dict(type='cartpole', import_names=['dizoo.classic_control.cartpole.envs.cartpole_env']),
+            env_manager=dict(type='base'),
+            policy=dict(type='dqn'),
+        ))
+    """))
+    out = subprocess.run(
+        [sys.executable, '-m', 'ding.entry.cli', '-m', 'serial', '-c', str(cfg), '-s', '0',
+         '--train-iter', '2'],
+        capture_output=True, text=True, timeout=300,
+    )
+    assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-2000:]
