@@ -14,4 +14,7 @@ from .serial_entry_variants2 import (
     serial_pipeline_guided_cost, serial_pipeline_td3_vae, serial_pipeline_onpolicy_ppg,
     serial_pipeline_bco, serial_pipeline_pc, trex_collecting_data, serial_pipeline_dreamer, serial_pipeline_plr, generate_seeds,
 )
-from .dist_entry import dist_prepare_config, dist_launch_coordinator, dist_launch_learner, dist_launch_collector
+from .dist_entry import (
+    dist_prepare_config, dist_launch_coordinator, dist_launch_learner, dist_launch_collector, dist_add_replicas,
+    dist_delete_replicas, dist_restart_replicas,
+)

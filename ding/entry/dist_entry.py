@@ -104,3 +104,40 @@ def dist_launch_coordinator(
     if collector is not None and started['collector']:
         _run(collector, {'name': 'collector_close_task'})
     return history
+
+
+def _operator_server_from_cfg(cfg: EasyDict):
+    """Build the DI-orchestrator client from cfg/env (reference
+    dist_entry.py:249-340 k8s replica commands)."""
+    from ding.utils import OperatorServer, get_operator_server_kwargs
+    kwargs = get_operator_server_kwargs(cfg.get('system', EasyDict({})))
+    server = OperatorServer(**kwargs)
+    server.set_worker_type('coordinator')
+    return server
+
+
+def dist_add_replicas(cfg: EasyDict, n_collectors: int = 0, n_learners: int = 0):
+    """Ask the orchestrator to scale UP collector/learner replicas."""
+    server = _operator_server_from_cfg(cfg)
+    ok, code, msg, data = server.post_replicas({'collectors': n_collectors, 'learners': n_learners})
+    if not ok:
+        raise RuntimeError(f'add replicas failed ({code}): {msg}')
+    return data
+
+
+def dist_delete_replicas(cfg: EasyDict, n_collectors: int = 0, n_learners: int = 0):
+    """Ask the orchestrator to scale DOWN collector/learner replicas."""
+    server = _operator_server_from_cfg(cfg)
+    ok, code, msg, data = server.delete_replicas(n_collectors, n_learners)
+    if not ok:
+        raise RuntimeError(f'delete replicas failed ({code}): {msg}')
+    return data
+
+
+def dist_restart_replicas(cfg: EasyDict, collectors=None, learners=None):
+    """Report failed replicas so the orchestrator restarts them."""
+    server = _operator_server_from_cfg(cfg)
+    ok, code, msg, data = server.post_replicas_failed(collectors or [], learners or [])
+    if not ok:
+        raise RuntimeError(f'restart replicas failed ({code}): {msg}')
+    return data

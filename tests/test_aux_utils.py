@@ -238,3 +238,39 @@ def test_k8s_env_kwargs_and_manifests(tmp_path, monkeypatch):
     manifest = launcher.create_manifest(output_path=str(tmp_path / 'orch.yaml'))
     assert 'di-operator' in manifest and 'di-server' in manifest
     assert (tmp_path / 'orch.yaml').exists()
+
+
+def test_dist_entry_replica_commands_against_stub():
+    """dist_{add,delete,restart}_replicas drive the orchestrator API."""
+    import threading
+    import time
+    from flask import Flask, jsonify, request as freq
+    from ding.entry import dist_add_replicas, dist_delete_replicas, dist_restart_replicas
+    from ding.utils import EasyDict
+    from ding.utils.misc_helpers import find_free_port
+
+    app = Flask('stub_orch2')
+    state = {'collectors': 0, 'failed': []}
+
+    @app.route('/v1alpha1/replicas', methods=['POST', 'DELETE'])
+    def replicas():
+        data = freq.get_json(silent=True) or {}
+        if freq.method == 'POST':
+            state['collectors'] += int(data.get('collectors', 0))
+        else:
+            state['collectors'] -= int(data['collectors']['replicas'])
+        return jsonify({'code': 0, 'message': 'success', 'data': {'collectors': state['collectors']}})
+
+    @app.route('/v1alpha1/replicas/failed', methods=['POST'])
+    def failed():
+        data = freq.get_json(silent=True) or {}
+        state['failed'] = data.get('collectors', [])
+        return jsonify({'code': 0, 'message': 'success', 'data': state['failed']})
+
+    port = find_free_port()
+    threading.Thread(target=lambda: app.run(host='127.0.0.1', port=port), daemon=True).start()
+    time.sleep(0.5)
+    cfg = EasyDict(dict(system=dict(system_addr=f'127.0.0.1:{port}')))
+    assert dist_add_replicas(cfg, n_collectors=3)['collectors'] == 3
+    assert dist_delete_replicas(cfg, n_collectors=1)['collectors'] == 2
+    assert dist_restart_replicas(cfg, collectors=['cl-0']) == ['cl-0']
