@@ -254,7 +254,7 @@ def test_gpu_per_buffer_hopper_scale():
     assert buf._count == 1_000_000
     # priorities skewed: heavy items must dominate samples
     hot = torch.arange(0, 1000, device='cuda')
-    buf.update_priority(hot, torch.full((1000, ), 100.0, device='cuda'))
+    buf.update_priority(hot, torch.full((1000, ), 1e6, device="cuda"))
     batch, idx, isw = buf.sample(256)
     assert batch['obs'].shape == (256, 11) and batch['obs'].is_cuda
     assert isw.shape == (256, )
