@@ -436,3 +436,20 @@ def test_image_classification_supervised():
             out = pol._forward_learn(batch)
         accs.append(out['acc'])
     assert accs[-1] > 0.5, f"classifier failed to learn: {accs}"
+
+
+def test_multi_discrete_dqn_policy():
+    """dizoo/common md_dqn: per-branch TD over a list of q heads."""
+    import torch
+    from dizoo.common.policy.md_policies import MultiDiscreteDQNPolicy
+    from ding.utils import EasyDict, deep_merge_dicts
+    cfg = EasyDict(deep_merge_dicts(MultiDiscreteDQNPolicy.default_config(), EasyDict(dict(
+        cuda=False, nstep=1,
+        model=dict(obs_shape=8, action_shape=[2, 3], encoder_hidden_size_list=[16, 16]),
+        learn=dict(batch_size=8, update_per_collect=1, learning_rate=1e-3),
+    ))))
+    pol = MultiDiscreteDQNPolicy(cfg, enable_field=['learn'])
+    data = [dict(obs=torch.randn(8), next_obs=torch.randn(8), action=torch.tensor([0, 2]),
+                 reward=torch.tensor([1.0]), done=False) for _ in range(8)]
+    losses = [pol._forward_learn([dict(d) for d in data])['total_loss'] for _ in range(5)]
+    assert losses[-1] < losses[0], losses
