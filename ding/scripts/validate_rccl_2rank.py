@@ -3,6 +3,11 @@ bucketed indicator all-reduce stays in sync and the trajectory shipper
 moves GPU tensors rank-to-rank. Run via torchrun; see bench_scale.sh for
 the full per-GPU scaling sweep."""
 import os
+import sys
+
+# torchrun children get the script's dir, not the repo root, on sys.path
+sys.path.insert(0, os.path.abspath(os.path.join(os.path.dirname(__file__), '..', '..')))
+
 import torch
 import torch.distributed as dist
 
