@@ -1,7 +1,11 @@
-"""Validate the RCCL codepath with 2 ranks sharing one GPU (CI helper):
-bucketed indicator all-reduce stays in sync and the trajectory shipper
-moves GPU tensors rank-to-rank. Run via torchrun; see bench_scale.sh for
-the full per-GPU scaling sweep."""
+"""Validate the RCCL codepath with 2 ranks (CI helper): bucketed indicator
+all-reduce stays in sync and the trajectory shipper moves GPU tensors
+rank-to-rank. Run via torchrun on a box with >= 2 GPUs — RCCL (like NCCL)
+refuses two ranks on the same device ("Duplicate GPU detected", verified on
+a 1-GPU MI355X box 2026-09-12), so each rank binds cuda:LOCAL_RANK. The
+same reducer/shipper logic is covered on CPU by the gloo world_size=2
+tests in tests/test_distributed.py. See bench_scale.sh for the full
+per-GPU scaling sweep."""
 import os
 import sys
 
@@ -14,7 +18,10 @@ import torch.distributed as dist
 
 def main():
     rank = int(os.environ['RANK'])
-    torch.cuda.set_device(0)
+    local = int(os.environ.get('LOCAL_RANK', rank))
+    if torch.cuda.device_count() < int(os.environ.get('WORLD_SIZE', 2)):
+        raise SystemExit('needs one GPU per rank: RCCL rejects two ranks on one device')
+    torch.cuda.set_device(local)
     dist.init_process_group('nccl')
     torch.manual_seed(100 + rank)
     # 1) bucketed reducer with indicator
