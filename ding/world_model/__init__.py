@@ -5,3 +5,4 @@ from .base_world_model import (
 from .mbpo import MBPOWorldModel, EnsembleModel, EnsembleFC
 from .ddppo import DDPPOWorldMode
 from .dreamer import DREAMERWorldModel, RSSM, ConvDecoder
+from .idm import InverseDynamicsModel
