@@ -174,3 +174,94 @@ def compile_config_parallel(cfg: EasyDict, create_cfg: EasyDict, system_cfg: Opt
     if system_cfg is not None:
         cfg.system = EasyDict(system_cfg)
     return cfg
+
+
+class Config:
+    """Dict+text config pair loaded from a python file (reference
+    ding/config/config.py Config:27). ``file_to_dict`` executes the file and
+    keeps every non-dunder module-level value in ``cfg_dict``."""
+
+    def __init__(self, cfg_dict: Optional[dict] = None, cfg_text: Optional[str] = None,
+                 filename: Optional[str] = None) -> None:
+        if cfg_dict is None:
+            cfg_dict = {}
+        if not isinstance(cfg_dict, dict):
+            raise TypeError(f"invalid type for cfg_dict: {type(cfg_dict)}")
+        self._cfg_dict = cfg_dict
+        if cfg_text:
+            self._text = cfg_text
+        elif filename:
+            with open(filename) as f:
+                self._text = f.read()
+        else:
+            self._text = ''
+        self._filename = filename
+
+    @staticmethod
+    def file_to_dict(filename: str) -> 'Config':
+        spec = importlib.util.spec_from_file_location('ding_user_config_full', filename)
+        module = importlib.util.module_from_spec(spec)
+        spec.loader.exec_module(module)
+        cfg_dict = {k: v for k, v in vars(module).items() if not k.startswith('__') and not callable(v)
+                    and not isinstance(v, type(importlib))}
+        return Config(cfg_dict, filename=filename)
+
+    @property
+    def cfg_dict(self) -> dict:
+        return self._cfg_dict
+
+
+def read_config_directly(path: str) -> dict:
+    """Whole-module config dict, no main/create splitting."""
+    assert path.endswith('.py'), f"invalid config file suffix: {path}"
+    return Config.file_to_dict(path).cfg_dict
+
+
+def read_config_with_system(path: str) -> Tuple[EasyDict, EasyDict, EasyDict]:
+    """(main_config, create_config, system_config) triple for parallel mode."""
+    cfg = read_config_directly(path)
+    for key in ('main_config', 'create_config', 'system_config'):
+        assert key in cfg, f"a '{key}' variable must be declared in {path}"
+    return EasyDict(cfg['main_config']), EasyDict(cfg['create_config']), EasyDict(cfg['system_config'])
+
+
+_DEFAULT_HOST, _DEFAULT_PORT = '127.0.0.1', 50515
+
+
+def parallel_transform(cfg: dict, coordinator_host: Optional[str] = None, learner_host=None,
+                       collector_host=None) -> EasyDict:
+    """Fill the system section (hosts/ports for coordinator, learners and
+    collectors) of a parallel-mode config (reference config/utils.py:195)."""
+    cfg = EasyDict(cfg)
+    coordinator_host = coordinator_host or _DEFAULT_HOST
+    learner_host = learner_host or [_DEFAULT_HOST]
+    collector_host = collector_host or [_DEFAULT_HOST]
+    if isinstance(learner_host, str):
+        learner_host = [learner_host]
+    if isinstance(collector_host, str):
+        collector_host = [collector_host]
+    system = cfg.get('system', EasyDict({}))
+    system.coordinator = EasyDict({'host': coordinator_host, 'port': _DEFAULT_PORT})
+    system.learners = EasyDict({
+        f'learner{i}': EasyDict({'host': h, 'port': _DEFAULT_PORT + 1 + i}) for i, h in enumerate(learner_host)
+    })
+    system.collectors = EasyDict({
+        f'collector{i}': EasyDict({'host': h, 'port': _DEFAULT_PORT + 101 + i}) for i, h in enumerate(collector_host)
+    })
+    cfg.system = system
+    return cfg
+
+
+def parallel_transform_slurm(cfg: dict, coordinator_host: Optional[str] = None, learner_node=None,
+                             collector_node=None) -> EasyDict:
+    """Slurm variant: node names resolve to hosts via node_to_host."""
+    from ding.utils import find_free_port_slurm
+    from ding.utils.misc_helpers import node_to_host
+    learner_host = [node_to_host(n) for n in (learner_node or [])] or None
+    collector_host = [node_to_host(n) for n in (collector_node or [])] or None
+    out = parallel_transform(cfg, coordinator_host, learner_host, collector_host)
+    # slurm boxes share nodes: replace fixed ports with job-derived free ones
+    for grp in (out.system.learners, out.system.collectors):
+        for v in grp.values():
+            v.port = find_free_port_slurm(v.host)
+    return out

@@ -132,3 +132,44 @@ def episode_to_transitions(data_path: str, expert_data_path: str, nstep: int) ->
         out.extend(list(get_nstep_return_data(deque(ep), nstep)))
     with open(expert_data_path, 'wb') as f:
         pickle.dump(out, f)
+
+
+def episode_to_transitions_filter(data_path: str, expert_data_path: str, nstep: int, min_episode_return: float) -> None:
+    """Like ``episode_to_transitions`` but drops episodes whose total return
+    is below ``min_episode_return`` (reference application_entry.py:258)."""
+    from collections import deque
+    from ding.rl_utils import get_nstep_return_data
+    with open(data_path, 'rb') as f:
+        episodes = pickle.load(f)
+    out = []
+    for ep in episodes:
+        ret = sum(float(step['reward'].sum() if hasattr(step['reward'], 'sum') else step['reward']) for step in ep)
+        if ret < min_episode_return:
+            continue
+        out.extend(list(get_nstep_return_data(deque(ep), nstep)))
+    with open(expert_data_path, 'wb') as f:
+        pickle.dump(out, f)
+
+
+def collect_episodic_demo_data_for_trex(
+    input_cfg,
+    seed: int,
+    collect_count: int,
+    rank: int,
+    save_cfg_path: str,
+    env_setting=None,
+    model=None,
+    state_dict=None,
+    state_dict_path: str = None,
+):
+    """TREX demo collection from one checkpoint stage: collect episodes with
+    the (partially trained) policy and write them under ``save_cfg_path``
+    tagged by checkpoint rank (reference application_entry_trex_collect_data.py:18)."""
+    import os
+    if state_dict is None and state_dict_path is not None:
+        state_dict = torch.load(state_dict_path, map_location='cpu')
+    os.makedirs(save_cfg_path, exist_ok=True)
+    expert_data_path = os.path.join(save_cfg_path, f'episodes_data_{rank}.pkl')
+    return collect_episodic_demo_data(
+        input_cfg, seed, collect_count, expert_data_path, env_setting=env_setting, model=model, state_dict=state_dict
+    )

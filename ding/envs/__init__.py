@@ -16,3 +16,5 @@ from .env_manager.subprocess_env_manager import (
 from .common.spaces import Discrete, Box, MultiDiscrete
 from .env_manager.env_supervisor import EnvSupervisor
 from .env.default_wrapper import get_default_wrappers
+from .env_manager import setup_ding_env_manager
+from . import gym_env

@@ -116,3 +116,6 @@ class LeagueMetricEnv:
 
 def get_elo(rating_a, rating_b, result):
     return EloCalculator.get_new_rating(rating_a, rating_b, result)
+
+
+get_elo_array = EloCalculator.get_new_rating_array

@@ -15,8 +15,10 @@ from .ppg import PPGPolicy, PPGOffPolicy
 from .acer import ACERPolicy
 from .dqfd import DQFDPolicy, PDQNPolicy, D4PGPolicy
 from .qmix import MADQNPolicy, CollaQPolicy, QTranPolicy
+QTRANPolicy = QTranPolicy  # reference spelling
+from .policy_factory import PolicyFactory, get_random_policy
 from .r2d2_variants import NGUPolicy, R2D3Policy, R2D2GTrXLPolicy, R2D2CollectTrajPolicy
-from .misc_policies import IBCPolicy, BCQPolicy, TD3VAEPolicy, PromptPGPolicy, PromptAWRPolicy, ProcedureCloningBFSPolicy
+from .misc_policies import ILPolicy, IBCPolicy, BCQPolicy, TD3VAEPolicy, PromptPGPolicy, PromptAWRPolicy, ProcedureCloningBFSPolicy
 from .happo import HAPPOPolicy
 from . import command_mode_policy_instance  # registers '<name>_command' variants
 from .mbpolicy import MBSACPolicy, STEVESACPolicy

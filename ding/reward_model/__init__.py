@@ -1,4 +1,4 @@
-from .base_reward_model import BaseRewardModel, create_reward_model
+from .base_reward_model import BaseRewardModel, create_reward_model, get_reward_model_cls
 from .exploration import RndRewardModel, ICMRewardModel
 from .imitation import (
     GailRewardModel, GuidedCostRewardModel, PwilRewardModel, RedRewardModel, PdeilRewardModel, TrexRewardModel,

@@ -55,6 +55,13 @@ class BaseRewardModel(ABC):
             self.reward_model.load_state_dict(d['model'])
 
 
+def get_reward_model_cls(cfg: EasyDict) -> type:
+    """Registry lookup without construction (reference base_reward_model.py:140)."""
+    from ding.utils import import_module
+    import_module(cfg.get('import_names', []))
+    return REWARD_MODEL_REGISTRY.get(cfg.type)
+
+
 def create_reward_model(cfg: EasyDict, device: str = 'cpu', tb_logger=None) -> BaseRewardModel:
     cfg = EasyDict(copy.deepcopy(cfg))
     if 'import_names' in cfg:

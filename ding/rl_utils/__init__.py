@@ -26,7 +26,7 @@ from .acer import acer_policy_error, acer_value_error, acer_trust_region_update
 from .ppg import ppg_data, ppg_joint_loss, ppg_joint_error
 from .happo import (
     happo_data, happo_policy_data, happo_value_data, happo_loss, happo_policy_loss, happo_info, happo_error,
-    happo_policy_error, happo_value_error, happo_error_continuous,
+    happo_policy_error, happo_value_error, happo_error_continuous, happo_policy_error_continuous,
 )
 from .grpo_rloo import grpo_policy_data, grpo_policy_error, grpo_info, rloo_policy_data, rloo_policy_error, rloo_info
 from .value_rescale import value_transform, value_inv_transform, symlog, inv_symlog

@@ -68,13 +68,14 @@ def happo_error(
     return happo_loss(pol.policy_loss, v_loss, pol.entropy_loss), info
 
 
-def happo_error_continuous(
+def happo_policy_error_continuous(
     data: namedtuple,
     clip_ratio: float = 0.2,
-    use_value_clip: bool = True,
     dual_clip: Optional[float] = None,
 ) -> Tuple[namedtuple, namedtuple]:
-    mu_sigma_new, mu_sigma_old, action, value_new, value_old, adv, return_, weight, factor = data
+    """Policy-only continuous HAPPO surrogate (factor-weighted).
+    Parity: reference ding/rl_utils/happo.py happo_policy_error_continuous:287."""
+    mu_sigma_new, mu_sigma_old, action, adv, weight, factor = data
     if weight is None:
         weight = torch.ones_like(adv)
     dist_new = _gaussian_dist(mu_sigma_new)
@@ -94,5 +95,18 @@ def happo_error_continuous(
     with torch.no_grad():
         approx_kl = (logp_old - logp_new).mean().item()
         clipfrac = ((ratio - 1.0).abs() > clip_ratio).float().mean().item()
+    return happo_policy_loss(policy_loss, entropy_loss), happo_info(approx_kl, clipfrac)
+
+
+def happo_error_continuous(
+    data: namedtuple,
+    clip_ratio: float = 0.2,
+    use_value_clip: bool = True,
+    dual_clip: Optional[float] = None,
+) -> Tuple[namedtuple, namedtuple]:
+    mu_sigma_new, mu_sigma_old, action, value_new, value_old, adv, return_, weight, factor = data
+    pol, info = happo_policy_error_continuous(
+        happo_policy_data(mu_sigma_new, mu_sigma_old, action, adv, weight, factor), clip_ratio, dual_clip
+    )
     v_loss = ppo_value_error(ppo_value_data(value_new, value_old, return_, weight), clip_ratio, use_value_clip)
-    return happo_loss(policy_loss, v_loss, entropy_loss), happo_info(approx_kl, clipfrac)
+    return happo_loss(pol.policy_loss, v_loss, pol.entropy_loss), info

@@ -223,3 +223,33 @@ class CudaFetcher:
 
     def __iter__(self):
         return self
+
+
+def tensor_to_list(item):
+    """Recursively convert tensors to (nested) python lists, leaving
+    non-tensor values untouched (reference data_helper.py:335)."""
+    if item is None:
+        return None
+    if isinstance(item, torch.Tensor):
+        return item.tolist()
+    if isinstance(item, dict):
+        return {k: tensor_to_list(v) for k, v in item.items()}
+    if isinstance(item, (list, tuple)):
+        return [tensor_to_list(v) for v in item]
+    if isinstance(item, (int, float, str, bool)):
+        return item
+    raise TypeError(f"not supported item type: {type(item)}")
+
+
+def get_null_data(template, num: int) -> list:
+    """Padding transitions stamped null+done with zeroed reward, used to pad
+    ragged episode batches (reference data_helper.py:696)."""
+    import copy as _copy
+    out = []
+    for _ in range(num):
+        data = _copy.deepcopy(template)
+        data['null'] = True
+        data['done'] = True
+        data['reward'].zero_()
+        out.append(data)
+    return out

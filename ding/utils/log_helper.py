@@ -170,3 +170,31 @@ def pretty_print(data: dict, direct_print: bool = True) -> str:
     if direct_print:
         print(text)
     return text
+
+
+class LoggerFactory:
+    """Factory view over build_logger: plain-logging creation plus the
+    tabulate helpers (reference log_helper.py:64)."""
+
+    @classmethod
+    def create_logger(cls, path, name: str = 'default', level=logging.INFO):
+        import os
+        os.makedirs(path, exist_ok=True)
+        logger = logging.getLogger(name)
+        logger.setLevel(level)
+        if not any(isinstance(h, logging.FileHandler) for h in logger.handlers):
+            fh = logging.FileHandler(os.path.join(path, f'{name}_logger.txt'))
+            fh.setFormatter(logging.Formatter('[%(asctime)s][%(levelname)s] %(message)s'))
+            logger.addHandler(fh)
+        return logger
+
+    @staticmethod
+    def get_tabulate_vars(variables: dict) -> str:
+        from tabulate import tabulate
+        return tabulate([[k, v] for k, v in variables.items()], headers=['Name', 'Value'], tablefmt='grid')
+
+    @staticmethod
+    def get_tabulate_vars_hor(variables: dict) -> str:
+        from tabulate import tabulate
+        keys, vals = list(variables.keys()), list(variables.values())
+        return tabulate([vals], headers=keys, tablefmt='grid')

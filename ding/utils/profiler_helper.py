@@ -1,5 +1,10 @@
 """cProfile wrapper. Parity: reference ding/utils/profiler_helper.py:12."""
 import atexit
+
+
+def register_profiler(write_profile, pr, folder_path):
+    """Flush the profile at interpreter exit (reference profiler_helper.py:8)."""
+    atexit.register(write_profile, pr, folder_path)
 import cProfile
 import os
 import pstats

@@ -86,6 +86,7 @@ MQ_REGISTRY = Registry("message_queue")
 STOP_VALUE_REGISTRY = Registry("stop_value")
 AGENT_REGISTRY = Registry("agent")
 HOOK_REGISTRY = Registry("hook")
+STOCHASTIC_OPTIMIZER_REGISTRY = Registry("stochastic_optimizer")
 
 REGISTRIES = {
     r.name: r
@@ -96,5 +97,6 @@ REGISTRIES = {
         LEARNER_REGISTRY, COMM_LEARNER_REGISTRY, COMM_COLLECTOR_REGISTRY, COMMANDER_REGISTRY,
         LEAGUE_REGISTRY, PLAYER_REGISTRY, MQ_REGISTRY, STOP_VALUE_REGISTRY, AGENT_REGISTRY,
         HOOK_REGISTRY,
+        STOCHASTIC_OPTIMIZER_REGISTRY,
     )
 }

@@ -182,6 +182,13 @@ class HybridWorldModel(DreamWorldModel, DynaWorldModel, ABC):
     pass
 
 
+def get_world_model_cls(cfg: EasyDict) -> type:
+    """Registry lookup without construction (reference base_world_model.py:17)."""
+    from ding.utils import import_module
+    import_module(cfg.get('import_names', []))
+    return WORLD_MODEL_REGISTRY.get(cfg.type)
+
+
 def create_world_model(cfg: EasyDict, env=None, tb_logger=None) -> WorldModel:
     cfg = EasyDict(copy.deepcopy(cfg))
     if 'import_names' in cfg:
