@@ -23,6 +23,18 @@ GPU_SUBSET = [
     ('dizoo.box2d.lunarlander.config.lunarlander_dqn_config', 'serial'),
     ('dizoo.dmc2gym.config.dmc2gym_sac_pixel_config', 'serial'),
     ('dizoo.mario.config.mario_dqn_config', 'serial'),
+    # breadth extension (round 2): distinct on-device codepaths — hybrid
+    # action PDQN, factor-sequential HAPPO, distributional d4pg, GTrXL
+    # attention unrolls, quantile fqf, offline decision transformer,
+    # recurrent NGU intrinsic stack, dreamer RSSM, multi-agent ATOC comm
+    ('dizoo.gym_hybrid.config.gym_hybrid_pdqn_config', 'serial'),
+    ('dizoo.petting_zoo.config.ptz_simple_spread_happo_config', 'onpolicy'),
+    ('dizoo.classic_control.pendulum.config.pendulum_d4pg_config', 'serial'),
+    ('dizoo.classic_control.cartpole.config.cartpole_r2d2_gtrxl_config', 'serial'),
+    ('dizoo.classic_control.cartpole.config.cartpole_fqf_config', 'serial'),
+    ('dizoo.d4rl.config.hopper_expert_dt_config', 'offline_dt'),
+    ('dizoo.dmc2gym.config.cartpole_balance_dreamer_config', 'dreamer'),
+    ('dizoo.petting_zoo.config.ptz_simple_spread_atoc_config', 'serial'),
 ]
 
 
