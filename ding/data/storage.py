@@ -35,3 +35,16 @@ class FileStorage(Storage):
     def load(self) -> Any:
         with open(self.path, "rb") as f:
             return pickle.load(f)
+
+
+class FileModelStorage(Storage):
+    """Model-checkpoint storage on the filesystem (reference
+    ding/data/storage/file.py:19)."""
+
+    def save(self, state_dict: object) -> None:
+        from ding.utils import save_file
+        save_file(self.path, state_dict)
+
+    def load(self) -> object:
+        from ding.utils import read_file
+        return read_file(self.path)

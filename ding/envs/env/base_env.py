@@ -81,3 +81,14 @@ def get_vec_env_setting(cfg: EasyDict, collect: bool = True, eval_: bool = True)
 def create_env(cfg: EasyDict) -> BaseEnv:
     import_module(cfg.get('import_names', []))
     return ENV_REGISTRY.build(cfg.type, cfg=cfg)
+
+
+def create_model_env(cfg: EasyDict):
+    """Build a model-env (world-model rollout env for mb-RL) from its registry
+    entry; remaining cfg keys go to the constructor (reference base_env.py:176)."""
+    import copy as _copy
+    cfg = _copy.deepcopy(cfg)
+    env_fn = get_env_cls(cfg)
+    cfg.pop('import_names', None)
+    cfg.pop('type', None)
+    return env_fn(**cfg)

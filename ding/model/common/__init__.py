@@ -3,6 +3,7 @@ from .encoder import (
     IMPALACnnResidualBlock, IMPALACnnDownStack,
 )
 from .head import (
+    independent_normal_dist,
     DiscreteHead, DistributionHead, RainbowHead, QRDQNHead, QuantileHead, FQFHead, DuelingHead, BranchingHead,
     StochasticDuelingHead, RegressionHead, ReparameterizationHead, PopArtVHead, AttentionPolicyHead, MultiHead,
     EnsembleHead, head_cls_map,

@@ -545,3 +545,13 @@ head_cls_map = {
     'ensemble': EnsembleHead,
     'sdn': StochasticDuelingHead,
 }
+
+
+def independent_normal_dist(logits) -> torch.distributions.Distribution:
+    """[mu, sigma] pair or {'mu','sigma'} dict -> Independent Normal over the
+    last dim (reference common/head.py:1440)."""
+    if isinstance(logits, (list, tuple)):
+        return torch.distributions.Independent(torch.distributions.Normal(*logits), 1)
+    if isinstance(logits, dict):
+        return torch.distributions.Independent(torch.distributions.Normal(logits['mu'], logits['sigma']), 1)
+    raise TypeError(f"invalid logits type: {type(logits)}")

@@ -1,5 +1,5 @@
 from .q_learning import DQN, BDQ, C51DQN, QRDQN, IQN, FQF, RainbowDQN, DRQN, GTrXLDQN
-from .vac import VAC
+from .vac import VAC, DREAMERVAC
 from .qac import ContinuousQAC, DiscreteQAC
 from .pg import PG
 from .qmix import QMix, Mixer

@@ -1,1 +1,1 @@
-from .model_wrappers import model_wrap, IModelWrapper, wrapper_name_map, TargetNetworkWrapper, HiddenStateWrapper
+from .model_wrappers import model_wrap, IModelWrapper, wrapper_name_map, register_wrapper, TargetNetworkWrapper, HiddenStateWrapper

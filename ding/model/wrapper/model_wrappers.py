@@ -517,3 +517,10 @@ def model_wrap(model: Union[nn.Module, IModelWrapper], wrapper_name: str = None,
     if wrapper_name not in wrapper_name_map:
         raise KeyError(f"unknown model wrapper: {wrapper_name}")
     return wrapper_name_map[wrapper_name](model, **kwargs)
+
+
+def register_wrapper(name: str, wrapper_type: type) -> None:
+    """Register a user wrapper so ``model_wrap(model, name)`` can build it
+    (reference model_wrappers.py:1014)."""
+    assert isinstance(name, str) and issubclass(wrapper_type, IModelWrapper)
+    wrapper_name_map[name] = wrapper_type
