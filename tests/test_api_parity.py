@@ -33,3 +33,31 @@ def test_package_surface_matches_reference(pkg):
     mine = importlib.import_module(f'ding.{pkg}')
     missing = [n for n in refs if not hasattr(mine, n)]
     assert missing == [], f"ding.{pkg} missing reference exports: {missing}"
+
+
+def _collect_star_resolved(pkg_init: str, ref_root: str):
+    """Names reachable through the reference's star imports, restricted to the
+    starred modules' own relative imports (stdlib/typing noise excluded)."""
+    tree = ast.parse(open(pkg_init).read())
+    names = []
+    for node in ast.walk(tree):
+        if isinstance(node, ast.ImportFrom):
+            if any(a.name == '*' for a in node.names):
+                base = os.path.join(ref_root, *(node.module or '').split('.'))
+                for cand in (base + '.py', os.path.join(base, '__init__.py')):
+                    if os.path.exists(cand):
+                        names += _collect_star_resolved(cand, os.path.dirname(cand))
+                        break
+            elif node.level >= 1:
+                names += [a.asname or a.name for a in node.names if a.name != '*']
+    return names
+
+
+@pytest.mark.skipif(not os.path.isdir(REFERENCE), reason="reference tree not present")
+@pytest.mark.parametrize('pkg', ['model', 'envs', 'data'])
+def test_star_import_surface_matches_reference(pkg):
+    refs = sorted(set(_collect_star_resolved(os.path.join(REFERENCE, pkg, '__init__.py'),
+                                             os.path.join(REFERENCE, pkg))))
+    mine = importlib.import_module(f'ding.{pkg}')
+    missing = [n for n in refs if not hasattr(mine, n)]
+    assert missing == [], f"ding.{pkg} missing star-resolved reference exports: {missing}"
