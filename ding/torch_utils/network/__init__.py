@@ -8,3 +8,8 @@ from .transformer import Transformer, TransformerLayer, Attention, ScaledDotProd
 from .gtrxl import GTrXL, GRUGatingUnit, PositionalEmbedding, AttentionXL, Memory
 from .scatter_connection import ScatterConnection
 from .popart import PopArt
+from .blocks_extra import (
+    Swish, Flatten, NearestUpsample, BilinearUpsample, SoftArgmax, GumbelSoftmax, GatingType, SumMerge, VectorMerge,
+    ResNet, resnet18,
+)
+from .nn_module import NoisyLinearLayer as NoiseLinearLayer  # reference spelling

@@ -205,3 +205,36 @@ def offline_data_save_type(exp_data: list, expert_data_path: str, data_type: str
         np.savez_compressed(path, **arrays)
     else:
         raise KeyError(data_type)
+
+
+@DATASET_REGISTRY.register('bco')
+class BCODataset(Dataset):
+    """(obs, action) pair dataset for Behavioral Cloning from Observation —
+    actions usually come from an inverse-dynamics model
+    (ding.world_model.InverseDynamicsModel). Parity: reference
+    ding/utils/data/dataset.py BCODataset:1136."""
+
+    def __init__(self, data=None):
+        if data is None:
+            raise ValueError('Dataset can not be empty!')
+        self._data = data
+
+    def __len__(self):
+        return len(self._data['obs'])
+
+    def __getitem__(self, idx):
+        return {k: v[idx] for k, v in self._data.items()}
+
+    @property
+    def obs(self):
+        return self._data['obs']
+
+    @property
+    def action(self):
+        return self._data['action']
+
+
+def hdf5_save(exp_data, expert_data_path: str) -> None:
+    """Save transitions as stacked arrays; uses the npz writer since h5py is
+    not shipped in this image (reference dataset.py:1494)."""
+    offline_data_save_type(exp_data, expert_data_path, data_type='hdf5')
