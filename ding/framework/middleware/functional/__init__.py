@@ -4,7 +4,7 @@ from .data_processor import (
     sqil_data_pusher,
 )
 from .trainer import trainer, multistep_trainer
-from .evaluator import interaction_evaluator, metric_evaluator, VectorEvalMonitor
+from .evaluator import interaction_evaluator, interaction_evaluator_ttorch, metric_evaluator, VectorEvalMonitor
 from .explorer import eps_greedy_handler, eps_greedy_masker
 from .advantage_estimator import gae_estimator, ppof_adv_estimator, montecarlo_return_estimator
 from .enhancer import reward_estimator, her_data_enhancer, nstep_reward_enhancer

@@ -115,3 +115,11 @@ def metric_evaluator(cfg: EasyDict, policy, dataloader, metric) -> Callable:
         ctx.eval_value = avg
 
     return _evaluate
+
+
+def interaction_evaluator_ttorch(cfg, policy, env, render: bool = False):
+    """Reference exposes a treetensor-typed evaluator variant
+    (functional/evaluator.py interaction_evaluator_ttorch); this build's
+    contexts are plain dict/tensor, so the one evaluator serves both —
+    kept as a named alias for API parity."""
+    return interaction_evaluator(cfg, policy, env, render=render)

@@ -8,3 +8,11 @@ from .adapter.learner_aggregator import LearnerAggregator
 from .collector.battle_collector import BattleSampleSerialCollector, BattleEpisodeSerialCollector, BattleInteractionSerialEvaluator
 from .collector.metric_serial_evaluator import MetricSerialEvaluator, IMetric
 from .coordinator.parallel_commander import BaseCommander, NaiveCommander, SoloCommander, OneVsOneCommander, create_parallel_commander
+from .comm import FlaskFileSystemLearner, FlaskFileSystemCollector
+from .legacy_parity import (
+    LearnerHook, register_learner_hook, add_learner_hook, merge_hooks, BaseCommLearner, BaseCommCollector,
+    create_comm_learner, create_comm_collector, VectorEvalMonitor, to_tensor_transitions, SequenceReplayBuffer,
+    BaseParallelCollector, ZerglingParallelCollector, MarineParallelCollector, NaiveCollector,
+    create_parallel_collector, get_parallel_collector_cls, get_parallel_commander_cls,
+)
+from .collector.interaction_serial_evaluator import InteractionSerialEvaluator as ISerialEvaluator

@@ -54,10 +54,15 @@ def _collect_star_resolved(pkg_init: str, ref_root: str):
 
 
 @pytest.mark.skipif(not os.path.isdir(REFERENCE), reason="reference tree not present")
-@pytest.mark.parametrize('pkg', ['model', 'envs', 'data'])
+@pytest.mark.parametrize(
+    'pkg', [
+        'model', 'envs', 'data', 'utils/data', 'worker', 'torch_utils/network', 'framework/middleware',
+        'envs/env_wrappers', 'envs/env_manager'
+    ]
+)
 def test_star_import_surface_matches_reference(pkg):
     refs = sorted(set(_collect_star_resolved(os.path.join(REFERENCE, pkg, '__init__.py'),
                                              os.path.join(REFERENCE, pkg))))
-    mine = importlib.import_module(f'ding.{pkg}')
+    mine = importlib.import_module('ding.' + pkg.replace('/', '.'))
     missing = [n for n in refs if not hasattr(mine, n)]
     assert missing == [], f"ding.{pkg} missing star-resolved reference exports: {missing}"
