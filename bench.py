@@ -212,6 +212,12 @@ def main():
     if use_gpu:
         torch.cuda.set_device(local_rank % max(1, torch.cuda.device_count()))
         device = f'cuda:{torch.cuda.current_device()}'
+        if world_size > 1 and 'MIOPEN_USER_DB_PATH' not in os.environ:
+            # per-rank find-db: 8 ranks sharing one sqlite user-db serialize
+            # (or corrupt) the find-mode writes during warmup
+            db = f'/tmp/miopen_udb_rank{local_rank}'
+            os.makedirs(db, exist_ok=True)
+            os.environ['MIOPEN_USER_DB_PATH'] = db
     else:
         device = 'cpu'
 
