@@ -25,6 +25,13 @@ def default_preprocess_learn(
     batches resident as stacked tensors instead of re-collating lists)."""
     if not isinstance(data, dict):
         data = default_collate(data)
+    for k in ('obs', 'next_obs'):
+        # frame-stack envs keep uint8 frames in the buffer (4x less HBM);
+        # models expect float at entry. nstep>1 rewrites next_obs from the
+        # (already float) obs stream, nstep==1 keeps the raw env frame — so
+        # cast here, once per batch.
+        if k in data and isinstance(data[k], torch.Tensor) and data[k].dtype == torch.uint8:
+            data[k] = data[k].float()
     if 'value_gamma' in data and isinstance(data['value_gamma'], list):
         data['value_gamma'] = torch.as_tensor(data['value_gamma'], dtype=torch.float32)
     if ignore_done:

@@ -1,0 +1,7 @@
+"""enduro impala (reference dizoo/atari/config/serial/enduro/enduro_impala_config.py;
+built by the shared factory — see atari_family.py)."""
+from dizoo.atari.config.serial.atari_family import build_atari_config
+
+main_config, create_config = build_atari_config('enduro', 'impala')
+enduro_impala_config = main_config
+enduro_impala_create_config = create_config

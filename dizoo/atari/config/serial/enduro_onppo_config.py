@@ -1,0 +1,7 @@
+"""enduro onppo (reference dizoo/atari/config/serial/enduro/enduro_onppo_config.py;
+built by the shared factory — see atari_family.py)."""
+from dizoo.atari.config.serial.atari_family import build_atari_config
+
+main_config, create_config = build_atari_config('enduro', 'onppo')
+enduro_onppo_config = main_config
+enduro_onppo_create_config = create_config

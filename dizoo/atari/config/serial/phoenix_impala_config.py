@@ -1,0 +1,7 @@
+"""phoenix impala (reference dizoo/atari/config/serial/phoenix/phoenix_impala_config.py;
+built by the shared factory — see atari_family.py)."""
+from dizoo.atari.config.serial.atari_family import build_atari_config
+
+main_config, create_config = build_atari_config('phoenix', 'impala')
+phoenix_impala_config = main_config
+phoenix_impala_create_config = create_config

@@ -1,0 +1,7 @@
+"""pong a2c (reference dizoo/atari/config/serial/pong/pong_a2c_config.py;
+built by the shared factory — see atari_family.py)."""
+from dizoo.atari.config.serial.atari_family import build_atari_config
+
+main_config, create_config = build_atari_config('pong', 'a2c')
+pong_a2c_config = main_config
+pong_a2c_create_config = create_config

@@ -1,0 +1,7 @@
+"""pong onppo (reference dizoo/atari/config/serial/pong/pong_onppo_config.py;
+built by the shared factory — see atari_family.py)."""
+from dizoo.atari.config.serial.atari_family import build_atari_config
+
+main_config, create_config = build_atari_config('pong', 'onppo')
+pong_onppo_config = main_config
+pong_onppo_create_config = create_config

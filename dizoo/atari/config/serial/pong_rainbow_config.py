@@ -1,0 +1,7 @@
+"""pong rainbow (reference dizoo/atari/config/serial/pong/pong_rainbow_config.py;
+built by the shared factory — see atari_family.py)."""
+from dizoo.atari.config.serial.atari_family import build_atari_config
+
+main_config, create_config = build_atari_config('pong', 'rainbow')
+pong_rainbow_config = main_config
+pong_rainbow_create_config = create_config

@@ -1,0 +1,7 @@
+"""qbert acer (reference dizoo/atari/config/serial/qbert/qbert_acer_config.py;
+built by the shared factory — see atari_family.py)."""
+from dizoo.atari.config.serial.atari_family import build_atari_config
+
+main_config, create_config = build_atari_config('qbert', 'acer')
+qbert_acer_config = main_config
+qbert_acer_create_config = create_config

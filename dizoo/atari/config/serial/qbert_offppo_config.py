@@ -1,0 +1,7 @@
+"""qbert offppo (reference dizoo/atari/config/serial/qbert/qbert_offppo_config.py;
+built by the shared factory — see atari_family.py)."""
+from dizoo.atari.config.serial.atari_family import build_atari_config
+
+main_config, create_config = build_atari_config('qbert', 'offppo')
+qbert_offppo_config = main_config
+qbert_offppo_create_config = create_config

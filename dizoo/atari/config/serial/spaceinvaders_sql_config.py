@@ -1,0 +1,7 @@
+"""spaceinvaders sql (reference dizoo/atari/config/serial/spaceinvaders/spaceinvaders_sql_config.py;
+built by the shared factory — see atari_family.py)."""
+from dizoo.atari.config.serial.atari_family import build_atari_config
+
+main_config, create_config = build_atari_config('spaceinvaders', 'sql')
+spaceinvaders_sql_config = main_config
+spaceinvaders_sql_create_config = create_config

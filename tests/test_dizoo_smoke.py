@@ -138,6 +138,10 @@ SMOKE_CONFIGS = [
     ('dizoo.gym_pybullet_drones.config.drone_hover_td3_config', 'serial'),
     ('dizoo.overcooked.config.overcooked_qmix_config', 'serial'),
     ('dizoo.atari.config.serial.pong_dqn_config', 'serial'),
+    ('dizoo.atari.config.serial.qbert_rainbow_config', 'serial'),
+    ('dizoo.atari.config.serial.spaceinvaders_offppo_config', 'serial'),
+    ('dizoo.atari.config.serial.qbert_impala_config', 'serial'),
+    ('dizoo.atari.config.serial.pong_mdqn_config', 'serial'),
     ('dizoo.mario.config.mario_dqn_config', 'serial'),
     ('dizoo.metadrive.config.metadrive_onppo_config', 'onpolicy'),
     ('dizoo.evogym.config.walker_ppo_config', 'onpolicy'),
