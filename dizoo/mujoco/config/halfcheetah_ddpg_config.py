@@ -1,0 +1,7 @@
+"""halfcheetah ddpg (reference dizoo/mujoco/config/halfcheetah_ddpg_config.py; built by the
+shared factory — see mujoco_family.py)."""
+from dizoo.mujoco.config.mujoco_family import build_mujoco_config
+
+main_config, create_config = build_mujoco_config('halfcheetah', 'ddpg')
+halfcheetah_ddpg_config = main_config
+halfcheetah_ddpg_create_config = create_config

@@ -1,0 +1,7 @@
+"""halfcheetah sac (reference dizoo/mujoco/config/halfcheetah_sac_config.py; built by the
+shared factory — see mujoco_family.py)."""
+from dizoo.mujoco.config.mujoco_family import build_mujoco_config
+
+main_config, create_config = build_mujoco_config('halfcheetah', 'sac')
+halfcheetah_sac_config = main_config
+halfcheetah_sac_create_config = create_config

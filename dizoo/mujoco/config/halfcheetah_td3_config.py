@@ -1,0 +1,7 @@
+"""halfcheetah td3 (reference dizoo/mujoco/config/halfcheetah_td3_config.py; built by the
+shared factory — see mujoco_family.py)."""
+from dizoo.mujoco.config.mujoco_family import build_mujoco_config
+
+main_config, create_config = build_mujoco_config('halfcheetah', 'td3')
+halfcheetah_td3_config = main_config
+halfcheetah_td3_create_config = create_config

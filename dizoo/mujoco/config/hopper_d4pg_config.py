@@ -1,0 +1,7 @@
+"""hopper d4pg (reference dizoo/mujoco/config/hopper_d4pg_config.py; built by the
+shared factory — see mujoco_family.py)."""
+from dizoo.mujoco.config.mujoco_family import build_mujoco_config
+
+main_config, create_config = build_mujoco_config('hopper', 'd4pg')
+hopper_d4pg_config = main_config
+hopper_d4pg_create_config = create_config

@@ -1,0 +1,7 @@
+"""hopper ddpg (reference dizoo/mujoco/config/hopper_ddpg_config.py; built by the
+shared factory — see mujoco_family.py)."""
+from dizoo.mujoco.config.mujoco_family import build_mujoco_config
+
+main_config, create_config = build_mujoco_config('hopper', 'ddpg')
+hopper_ddpg_config = main_config
+hopper_ddpg_create_config = create_config

@@ -1,0 +1,7 @@
+"""humanoid sac (reference dizoo/mujoco/config/humanoid_sac_config.py; built by the
+shared factory — see mujoco_family.py)."""
+from dizoo.mujoco.config.mujoco_family import build_mujoco_config
+
+main_config, create_config = build_mujoco_config('humanoid', 'sac')
+humanoid_sac_config = main_config
+humanoid_sac_create_config = create_config

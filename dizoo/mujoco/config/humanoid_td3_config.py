@@ -1,0 +1,7 @@
+"""humanoid td3 (reference dizoo/mujoco/config/humanoid_td3_config.py; built by the
+shared factory — see mujoco_family.py)."""
+from dizoo.mujoco.config.mujoco_family import build_mujoco_config
+
+main_config, create_config = build_mujoco_config('humanoid', 'td3')
+humanoid_td3_config = main_config
+humanoid_td3_create_config = create_config

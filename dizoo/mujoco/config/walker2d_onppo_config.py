@@ -1,0 +1,7 @@
+"""walker2d onppo (reference dizoo/mujoco/config/walker2d_onppo_config.py; built by the
+shared factory — see mujoco_family.py)."""
+from dizoo.mujoco.config.mujoco_family import build_mujoco_config
+
+main_config, create_config = build_mujoco_config('walker2d', 'onppo')
+walker2d_onppo_config = main_config
+walker2d_onppo_create_config = create_config

@@ -18,6 +18,8 @@ _PRESETS = {
     'Hopper-v3': dict(obs=11, act=3),
     'HalfCheetah-v3': dict(obs=17, act=6),
     'Walker2d-v3': dict(obs=17, act=6),
+    'Ant-v3': dict(obs=111, act=8),
+    'Humanoid-v3': dict(obs=376, act=17),
 }
 
 

@@ -116,6 +116,10 @@ SMOKE_CONFIGS = [
     ('dizoo.classic_control.pendulum.config.pendulum_mbsac_ddppo_config', 'dream'),
     ('dizoo.classic_control.pendulum.config.pendulum_stevesac_ddppo_config', 'dream'),
     ('dizoo.mujoco.config.hopper_bdq_config', 'serial'),
+    ('dizoo.mujoco.config.walker2d_td3_config', 'serial'),
+    ('dizoo.mujoco.config.hopper_d4pg_config', 'serial'),
+    ('dizoo.mujoco.config.ant_sac_config', 'serial'),
+    ('dizoo.mujoco.config.humanoid_onppo_config', 'onpolicy'),
     ('dizoo.maze.config.maze_pc_config', 'pc'),
     ('dizoo.dmc2gym.config.cartpole_balance_dreamer_config', 'dreamer'),
     ('dizoo.tabmwp.config.tabmwp_prompt_pg_config', 'onpolicy'),
