@@ -53,12 +53,25 @@ class DiscreteMAQAC(nn.Module):
         assert mode in self.mode, mode
         return getattr(self, mode)(inputs)
 
+    @staticmethod
+    def _obs(inputs: Dict) -> Dict:
+        # accept both {'obs': {...}} (learn batches) and the bare obs dict
+        # (collect/eval forward)
+        return inputs['obs'] if 'obs' in inputs else inputs
+
     def compute_actor(self, inputs: Dict) -> Dict:
-        x = self.actor(inputs['obs']['agent_state'])
-        return {'logit': x['logit'], 'action_mask': inputs['obs'].get('action_mask')}
+        obs = self._obs(inputs)
+        x = self.actor(obs['agent_state'])
+        return {'logit': x['logit'], 'action_mask': obs.get('action_mask')}
 
     def compute_critic(self, inputs: Dict) -> Dict:
-        gs = inputs['obs']['global_state']
+        obs = self._obs(inputs)
+        gs = obs['global_state']
+        if 'agent_state' in obs and gs.dim() == obs['agent_state'].dim() - 1:
+            # shared global state: broadcast across the agent dim so the
+            # critic emits per-agent Q ([B, A, N])
+            A = obs['agent_state'].shape[-2]
+            gs = gs.unsqueeze(-2).expand(*gs.shape[:-1], A, gs.shape[-1])
         if self.twin_critic:
             return {'q_value': [m(gs)['logit'] for m in self.critic]}
         return {'q_value': self.critic(gs)['logit']}

@@ -1,0 +1,7 @@
+"""SMAC 25m masac (reference dizoo/smac/config/smac_25m_masac_config.py; built by the
+shared factory — see smac_family.py)."""
+from dizoo.smac.config.smac_family import build_smac_config
+
+main_config, create_config = build_smac_config('25m', 'masac')
+smac_25m_masac_config = main_config
+smac_25m_masac_create_config = create_config

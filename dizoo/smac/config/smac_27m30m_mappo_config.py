@@ -1,0 +1,7 @@
+"""SMAC 27m30m mappo (reference dizoo/smac/config/smac_27m30m_mappo_config.py; built by the
+shared factory — see smac_family.py)."""
+from dizoo.smac.config.smac_family import build_smac_config
+
+main_config, create_config = build_smac_config('27m30m', 'mappo')
+smac_27m30m_mappo_config = main_config
+smac_27m30m_mappo_create_config = create_config

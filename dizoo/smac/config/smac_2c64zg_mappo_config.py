@@ -1,0 +1,7 @@
+"""SMAC 2c64zg mappo (reference dizoo/smac/config/smac_2c64zg_mappo_config.py; built by the
+shared factory — see smac_family.py)."""
+from dizoo.smac.config.smac_family import build_smac_config
+
+main_config, create_config = build_smac_config('2c64zg', 'mappo')
+smac_2c64zg_mappo_config = main_config
+smac_2c64zg_mappo_create_config = create_config

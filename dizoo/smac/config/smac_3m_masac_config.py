@@ -1,0 +1,7 @@
+"""SMAC 3m masac (reference dizoo/smac/config/smac_3m_masac_config.py; built by the
+shared factory — see smac_family.py)."""
+from dizoo.smac.config.smac_family import build_smac_config
+
+main_config, create_config = build_smac_config('3m', 'masac')
+smac_3m_masac_config = main_config
+smac_3m_masac_create_config = create_config

@@ -1,0 +1,7 @@
+"""SMAC 5m6m madqn (reference dizoo/smac/config/smac_5m6m_madqn_config.py; built by the
+shared factory — see smac_family.py)."""
+from dizoo.smac.config.smac_family import build_smac_config
+
+main_config, create_config = build_smac_config('5m6m', 'madqn')
+smac_5m6m_madqn_config = main_config
+smac_5m6m_madqn_create_config = create_config

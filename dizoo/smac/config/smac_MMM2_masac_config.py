@@ -1,0 +1,7 @@
+"""SMAC MMM2 masac (reference dizoo/smac/config/smac_MMM2_masac_config.py; built by the
+shared factory — see smac_family.py)."""
+from dizoo.smac.config.smac_family import build_smac_config
+
+main_config, create_config = build_smac_config('MMM2', 'masac')
+smac_MMM2_masac_config = main_config
+smac_MMM2_masac_create_config = create_config

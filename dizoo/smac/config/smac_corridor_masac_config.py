@@ -1,0 +1,7 @@
+"""SMAC corridor masac (reference dizoo/smac/config/smac_corridor_masac_config.py; built by the
+shared factory — see smac_family.py)."""
+from dizoo.smac.config.smac_family import build_smac_config
+
+main_config, create_config = build_smac_config('corridor', 'masac')
+smac_corridor_masac_config = main_config
+smac_corridor_masac_create_config = create_config

@@ -61,6 +61,12 @@ SMOKE_CONFIGS = [
     ('dizoo.gym_hybrid.config.gym_hybrid_hppo_config', 'onpolicy'),
     ('dizoo.gym_hybrid.config.gym_hybrid_ddpg_config', 'serial'),
     ('dizoo.petting_zoo.config.ptz_simple_spread_qmix_config', 'serial'),
+    ('dizoo.smac.config.smac_3s5z_qmix_config', 'serial'),
+    ('dizoo.smac.config.smac_MMM_qtran_config', 'serial'),
+    ('dizoo.smac.config.smac_5m6m_wqmix_config', 'serial'),
+    ('dizoo.smac.config.smac_10m11m_mappo_config', 'onpolicy'),
+    ('dizoo.smac.config.smac_3s5z_masac_config', 'serial'),
+    ('dizoo.smac.config.smac_MMM2_madqn_config', 'serial'),
     ('dizoo.petting_zoo.config.ptz_simple_spread_vdn_config', 'serial'),
     ('dizoo.petting_zoo.config.ptz_simple_spread_wqmix_config', 'serial'),
     ('dizoo.petting_zoo.config.ptz_simple_spread_qtran_config', 'serial'),
