@@ -100,6 +100,13 @@ SMOKE_CONFIGS = [
     ('dizoo.dmc2gym.config.dmc2gym_ppo_config', 'onpolicy'),
     ('dizoo.cliffwalking.config.cliffwalking_dqn_config', 'serial'),
     ('dizoo.classic_control.cartpole.config.cartpole_c51_config', 'serial'),
+    ('dizoo.classic_control.cartpole.config.cartpole_ppo_config', 'onpolicy'),
+    ('dizoo.classic_control.cartpole.config.cartpole_a2c_config', 'onpolicy'),
+    ('dizoo.classic_control.cartpole.config.cartpole_impala_config', 'serial'),
+    ('dizoo.classic_control.cartpole.config.cartpole_sac_config', 'serial'),
+    ('dizoo.classic_control.cartpole.config.cartpole_r2d2_config', 'serial'),
+    ('dizoo.classic_control.cartpole.config.cartpole_sql_config', 'serial'),
+    ('dizoo.classic_control.cartpole.config.cartpole_ppo_offpolicy_config', 'serial'),
     ('dizoo.classic_control.cartpole.config.cartpole_qrdqn_config', 'serial'),
     ('dizoo.classic_control.cartpole.config.cartpole_iqn_config', 'serial'),
     ('dizoo.classic_control.cartpole.config.cartpole_fqf_config', 'serial'),
