@@ -43,6 +43,9 @@ def _shrink(main_cfg, create_cfg):
 # (config module, pipeline entry)
 SMOKE_CONFIGS = [
     ('dizoo.box2d.lunarlander.config.lunarlander_dqn_config', 'serial'),
+    ('dizoo.box2d.lunarlander.config.lunarlander_c51_config', 'serial'),
+    ('dizoo.box2d.lunarlander.config.lunarlander_rainbow_config', 'serial'),
+    ('dizoo.box2d.lunarlander.config.lunarlander_r2d2_config', 'serial'),
     ('dizoo.box2d.lunarlander.config.lunarlander_ppo_config', 'onpolicy'),
     ('dizoo.box2d.lunarlander.config.lunarlander_a2c_config', 'onpolicy'),
     ('dizoo.box2d.lunarlander.config.lunarlander_impala_config', 'serial'),
