@@ -35,6 +35,11 @@ GPU_SUBSET = [
     ('dizoo.d4rl.config.hopper_expert_dt_config', 'offline_dt'),
     ('dizoo.dmc2gym.config.cartpole_balance_dreamer_config', 'dreamer'),
     ('dizoo.petting_zoo.config.ptz_simple_spread_atoc_config', 'serial'),
+    # round-2 late families: factory-built atari/smac/mujoco configs
+    ('dizoo.atari.config.serial.qbert_rainbow_config', 'serial'),
+    ('dizoo.atari.config.serial.pong_mdqn_config', 'serial'),
+    ('dizoo.smac.config.smac_3s5z_masac_config', 'serial'),
+    ('dizoo.mujoco.config.hopper_d4pg_config', 'serial'),
 ]
 
 
