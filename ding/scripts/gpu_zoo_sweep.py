@@ -18,7 +18,7 @@ CASES = [
     ('dizoo.atari.config.serial.pong_stdim_config', 'serial'),
     ('dizoo.smac.config.smac_MMM_qtran_config', 'serial'),
     ('dizoo.smac.config.smac_25m_mappo_config', 'onpolicy'),
-    ('dizoo.smac.config.smac_2c64zg_collaq_config', 'serial'),
+    ('dizoo.smac.config.smac_3s5z_collaq_config', 'serial'),
     ('dizoo.mujoco.config.walker2d_td3_config', 'serial'),
     ('dizoo.mujoco.config.humanoid_onppo_config', 'onpolicy'),
     ('dizoo.mujoco.config.halfcheetah_bdq_config', 'serial'),
