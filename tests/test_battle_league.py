@@ -204,3 +204,18 @@ def test_trueskill_rating_math():
     assert a2.sigma < a.sigma, "uncertainty must shrink after a match"
     # exposure = mu - 3 sigma grows for the winner
     assert a2.exposure > a.exposure
+
+
+def test_selfplay_demo_ppo_main_runs():
+    """Reference dizoo/league_demo/selfplay_demo_ppo_main.py analog."""
+    from dizoo.league_demo.selfplay_demo_ppo_main import main
+    pols = main(max_train_iter=4)
+    assert len(pols) == 2
+
+
+def test_league_demo_ppo_main_snapshots():
+    """Reference dizoo/league_demo/league_demo_ppo_main.py analog: the league
+    loop must produce at least one historical snapshot."""
+    from dizoo.league_demo.league_demo_ppo_main import main
+    _, league, payoff = main(max_train_iter=25)
+    assert len(league.historical_players) >= 1
