@@ -1,0 +1,12 @@
+"""c51_nstep.py middleware example (reference ding/example/c51_nstep.py)."""
+from ding.policy import C51Policy
+from .common import cartpole_envs, offpolicy_main
+
+
+def main(max_step: int = 1000):
+    return offpolicy_main('dizoo.classic_control.cartpole.config.cartpole_c51_config', C51Policy, envs_fn=cartpole_envs, max_step=max_step,
+                          use_nstep=True, use_eps=True)
+
+
+if __name__ == '__main__':
+    main()

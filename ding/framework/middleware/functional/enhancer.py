@@ -49,7 +49,7 @@ def nstep_reward_enhancer(cfg: EasyDict) -> Callable:
 
     def _enhance(ctx: OnlineRLContext):
         nstep = cfg.policy.nstep
-        gamma = cfg.policy.discount_factor
+        gamma = cfg.policy.get('discount_factor', cfg.policy.learn.get('discount_factor', 0.99))
         from collections import deque
         # per-env slices to avoid crossing trajectory boundaries
         start = 0
