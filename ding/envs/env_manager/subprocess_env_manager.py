@@ -27,14 +27,25 @@ logger = logging.getLogger('ding')
 def _worker_loop(child_conn, env_fn, shm_buffer, auto_reset: bool):
     """Child process: serve reset/step/seed/close/method commands."""
     env = env_fn()
+
+    def _try_shm(obs):
+        """Best-effort shm transport: envs whose obs doesn't match the
+        declared observation_space (e.g. MARL dict obs) fall back to the
+        pipe instead of crashing the worker."""
+        if shm_buffer is None:
+            return False
+        try:
+            shm_buffer.fill(np.ascontiguousarray(obs))
+            return True
+        except Exception:
+            return False
     try:
         while True:
             cmd, payload = child_conn.recv()
             try:
                 if cmd == 'reset':
                     obs = env.reset(**payload)
-                    if shm_buffer is not None:
-                        shm_buffer.fill(np.ascontiguousarray(obs))
+                    if _try_shm(obs):
                         child_conn.send(('ok', ('shm', )))
                     else:
                         child_conn.send(('ok', obs))
@@ -46,11 +57,11 @@ def _worker_loop(child_conn, env_fn, shm_buffer, auto_reset: bool):
                         new_obs = env.reset()
                     else:
                         new_obs = None
-                    if shm_buffer is not None:
-                        shm_buffer.fill(np.ascontiguousarray(new_obs if (done and auto_reset) else obs))
+                    out_obs = new_obs if (done and auto_reset) else obs
+                    if _try_shm(out_obs):
                         child_conn.send(('ok', ('shm', ts.reward, done, ts.info)))
                     else:
-                        child_conn.send(('ok', (new_obs if (done and auto_reset) else obs, ts.reward, done, ts.info)))
+                        child_conn.send(('ok', (out_obs, ts.reward, done, ts.info)))
                 elif cmd == 'seed':
                     env.seed(*payload)
                     child_conn.send(('ok', None))

@@ -1,0 +1,15 @@
+"""edac.py example (reference ding/example/edac.py): runs the walker2d_medium_edac_config
+config through serial_pipeline_offline.
+offline: generate the dataset first (dizoo/d4rl/generate.py)"""
+from ding.entry import serial_pipeline_offline
+
+
+def main(max_train_iter: int = 100, seed: int = 0):
+    from dizoo.d4rl.config.walker2d_medium_edac_config import main_config, create_config
+    import copy
+    return serial_pipeline_offline((copy.deepcopy(main_config), copy.deepcopy(create_config)), seed=seed,
+                   max_train_iter=max_train_iter)
+
+
+if __name__ == '__main__':
+    main()

@@ -1,0 +1,14 @@
+"""masac.py example (reference ding/example/masac.py): runs the smac_3s5z_masac_config
+config through serial_pipeline."""
+from ding.entry import serial_pipeline
+
+
+def main(max_train_iter: int = 100, seed: int = 0):
+    from dizoo.smac.config.smac_3s5z_masac_config import main_config, create_config
+    import copy
+    return serial_pipeline((copy.deepcopy(main_config), copy.deepcopy(create_config)), seed=seed,
+                   max_train_iter=max_train_iter)
+
+
+if __name__ == '__main__':
+    main()

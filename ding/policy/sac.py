@@ -295,6 +295,12 @@ class DiscreteSACPolicy(SACPolicy):
             if self._twin_critic:
                 next_q = torch.min(next_q[0], next_q[1])
             target_v = (next_pi * (next_q - self._alpha * next_logpi)).sum(-1)
+            # multi-agent: target_v is [B, A] while reward/done are [B] —
+            # broadcast per agent (shared team reward / episode done)
+            while reward.dim() < target_v.dim():
+                reward = reward.unsqueeze(-1)
+            while done.dim() < target_v.dim():
+                done = done.unsqueeze(-1)
             target_q = reward + self._gamma * (1 - done) * target_v
         act = data['action'].long()
         if self._twin_critic:
