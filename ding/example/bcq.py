@@ -7,7 +7,9 @@ from ding.entry import serial_pipeline_offline
 def main(max_train_iter: int = 100, seed: int = 0):
     from dizoo.d4rl.config.hopper_medium_bcq_config import main_config, create_config
     import copy
-    return serial_pipeline_offline((copy.deepcopy(main_config), copy.deepcopy(create_config)), seed=seed,
+    m = copy.deepcopy(main_config)
+    m.exp_name = 'exp/example_bcq'
+    return serial_pipeline_offline((m, copy.deepcopy(create_config)), seed=seed,
                    max_train_iter=max_train_iter)
 
 

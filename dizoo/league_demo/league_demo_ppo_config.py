@@ -6,7 +6,7 @@ from ding.utils import EasyDict
 from dizoo.league_demo.selfplay_demo_ppo_config import selfplay_demo_ppo_config
 
 league_demo_ppo_config = EasyDict(dict(
-    exp_name='league_demo_ppo',
+    exp_name='exp/league_demo_ppo',
     env=dict(selfplay_demo_ppo_config.env),
     policy=dict(selfplay_demo_ppo_config.policy),
     league=dict(

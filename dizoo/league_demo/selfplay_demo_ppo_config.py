@@ -4,7 +4,7 @@ other; see selfplay_demo_ppo_main.py for the runnable loop."""
 from ding.utils import EasyDict
 
 selfplay_demo_ppo_config = EasyDict(dict(
-    exp_name='selfplay_demo_ppo',
+    exp_name='exp/selfplay_demo_ppo',
     env=dict(
         env_type='zero_sum',
         repeat_count=4,
