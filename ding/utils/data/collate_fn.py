@@ -72,8 +72,12 @@ def timestep_collate(batch: List[Dict[str, Any]]) -> Dict[str, Any]:
     elem = batch[0]
     prev_state = None
     if 'prev_state' in elem:
-        # [B][T] -> [T][B]
-        prev_state = list(zip(*[d.pop('prev_state') for d in batch]))
+        # [B][T] -> [T][B]. Do NOT pop in place: buffers hand out references
+        # to stored samples, and a destructive pop breaks re-sampling the
+        # same transition (update_per_collect > buffer turnover).
+        prev_state = list(zip(*[d['prev_state'] for d in batch]))
+        batch = [{k: v for k, v in d.items() if k != 'prev_state'} for d in batch]
+        elem = batch[0]
     def _stack_time(vals):
         """vals: B samples, each a T-list (or [T,...] tensor) -> [T, B, ...]."""
         if isinstance(vals[0], dict):

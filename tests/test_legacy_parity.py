@@ -129,3 +129,21 @@ def test_policy_factory_random_forward():
     assert set(out.keys()) == {0, 3}
     for v in out.values():
         assert 0 <= int(v['action']) < 5
+
+
+def test_timestep_collate_nondestructive():
+    """Re-collating the same stored samples must work: buffers hand out
+    references, so collate must not pop 'prev_state' in place (bug found by
+    the on-device QMIX learning check)."""
+    from ding.utils.data.collate_fn import timestep_collate
+    batch = [
+        {
+            'obs': [torch.randn(3) for _ in range(4)],
+            'action': [torch.zeros(1, dtype=torch.long) for _ in range(4)],
+            'prev_state': [None] * 4,
+        } for _ in range(2)
+    ]
+    out1 = timestep_collate(batch)
+    out2 = timestep_collate(batch)  # second pass over the SAME dicts
+    assert 'prev_state' in batch[0], "collate must not mutate stored samples"
+    assert out1['obs'].shape == out2['obs'].shape == (4, 2, 3)
